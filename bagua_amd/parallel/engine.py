@@ -1,0 +1,397 @@
+"""BaguaDistributedDataParallel — the training-loop engine.
+
+Re-design of the reference engine
+(bagua/torch_api/data_parallel/bagua_distributed.py:28-505) for current
+PyTorch on ROCm:
+
+* per-parameter hooks use ``register_post_accumulate_grad_hook`` (instead
+  of the grad_fn.next_functions expand-trick of old torch);
+* the post-backward callback still goes through
+  ``Variable._execution_engine.queue_callback`` so it runs when autograd
+  drains (reference: bagua_distributed.py:448-452);
+* bucketing defaults to local greedy size-based packing (the reference
+  asked the autotune service even at level 0 —
+  autotune_task_manager.py:85-119); with BAGUA_AUTOTUNE=1 the HTTP service
+  drives re-bucketing every 100 iterations.
+"""
+
+import logging
+import time
+from typing import Callable, Dict, List, Optional
+
+import torch
+from torch.autograd import Variable
+
+from .. import env
+from ..bucket import BaguaBucket
+from ..communication import (
+    BaguaProcessGroup,
+    _get_default_group,
+    broadcast_coalesced,
+    broadcast_object,
+    get_backend,
+)
+from ..defines import BaguaHyperparameter, TensorDeclaration, TensorDtype, dtype_bytes
+from ..tensor import BaguaTensor, ensure_bagua_tensor
+from .algorithms.base import Algorithm
+
+logger = logging.getLogger(__name__)
+
+_TORCH_DTYPE_MAP = {
+    torch.float32: TensorDtype.F32,
+    torch.float16: TensorDtype.F16,
+    torch.bfloat16: TensorDtype.BF16,
+    torch.uint8: TensorDtype.U8,
+    torch.int64: TensorDtype.I64,
+}
+
+
+def split_tensors_into_groups(
+    tensors: List[BaguaTensor], bucket_bytes: int
+) -> List[List[BaguaTensor]]:
+    """Greedy size-based packing, grouped by dtype, preserving order
+    (reference: autotune_task_manager.py:85-119)."""
+    groups: List[List[BaguaTensor]] = []
+    cur: List[BaguaTensor] = []
+    cur_bytes = 0
+    cur_dtype = None
+    for t in tensors:
+        eff = t.tensor()
+        nb = eff.numel() * eff.element_size()
+        if cur and (cur_dtype != eff.dtype or cur_bytes + nb > bucket_bytes):
+            groups.append(cur)
+            cur, cur_bytes = [], 0
+        cur.append(t)
+        cur_bytes += nb
+        cur_dtype = eff.dtype
+    if cur:
+        groups.append(cur)
+    return groups
+
+
+class BaguaDistributedDataParallel:
+    def __init__(
+        self,
+        module: torch.nn.Module,
+        optimizers: List[torch.optim.Optimizer],
+        algorithm: Algorithm,
+        process_group: Optional[BaguaProcessGroup] = None,
+        bagua_module_name: Optional[str] = None,
+        gradient_as_bucket_view: bool = True,
+        find_unused_parameters: bool = False,
+    ):
+        self.module = module
+        self.bagua_optimizers = list(optimizers)
+        self.bagua_algorithm = algorithm.reify(
+            process_group or _get_default_group())
+        self.process_group = process_group or _get_default_group()
+        if bagua_module_name is None:
+            bagua_module_name = "bagua_module_{}_{}".format(
+                module.__class__.__name__, id(module))
+        self.bagua_module_name = bagua_module_name
+        self.gradient_as_bucket_view = gradient_as_bucket_view
+        self.find_unused_parameters = find_unused_parameters
+
+        self.bagua_train_step_counter = 0
+        self.bagua_buckets: List[BaguaBucket] = []
+        self._bagua_tensor_map: Dict[str, BaguaTensor] = {}
+        self.require_backward_grad_sync = True
+        self.parameters_to_ignore = list(
+            getattr(module, "_bagua_params_and_buffers_to_ignore", []))
+
+        self.bagua_backend = get_backend(self.bagua_module_name)
+
+        # speed metrics for autotune scoring
+        # (reference: bagua_distributed.py:113-131)
+        self._speed_metrics_start = None
+        self._speed_samples = []
+
+        self._autotune_client = None
+        self._autotune_completed = env.get_autotune_level() == 0
+        self._bagua_autograd_hook_handles = []
+        self._hooks_fired_this_iter = False
+
+        self._install_forward_pre_hooks()
+        self._bagua_init_algorithm()
+
+    # ------------------------------------------------------------------
+    # parameter discovery
+    # ------------------------------------------------------------------
+    def bagua_build_params(self):
+        """Deduplicated, trainable (name, param) list; skips ignored and
+        MoE expert parameters (reference: bagua_distributed.py:155-212)."""
+        from .moe.utils import is_moe_param
+
+        modules_and_parameters = [
+            (name, param)
+            for name, param in self.module.named_parameters()
+            if param.requires_grad
+            and name not in self.parameters_to_ignore
+            and not is_moe_param(param)
+        ]
+        seen = set()
+        out = []
+        for name, param in modules_and_parameters:
+            if param.is_sparse:
+                raise ValueError("sparse parameters are not supported")
+            if id(param) in seen:
+                continue
+            seen.add(id(param))
+            out.append((name, param))
+        return out
+
+    def ensure_bagua_tensor(self, param, name, getter_closure=None,
+                            setter_closure=None) -> BaguaTensor:
+        t = ensure_bagua_tensor(param, name, getter_closure, setter_closure)
+        self._bagua_tensor_map[name] = t
+        return t
+
+    # ------------------------------------------------------------------
+    # state broadcast at init (reference: bagua_distributed.py:229-323)
+    # ------------------------------------------------------------------
+    def _bagua_broadcast_parameters(self):
+        comm = self.process_group.get_global_communicator()
+        module_states = []
+        for _, p in sorted(self.module.state_dict().items()):
+            if isinstance(p, torch.Tensor) and p.numel() > 0 \
+                    and p.dtype.is_floating_point:
+                module_states.append(p.data)
+            elif isinstance(p, torch.Tensor) and p.numel() > 0:
+                module_states.append(p.data)
+        if module_states:
+            broadcast_coalesced(module_states, src=0, comm=comm)
+        for opt in self.bagua_optimizers:
+            self._bagua_broadcast_optimizer_state(opt, comm)
+        if torch.cuda.is_available():
+            torch.cuda.current_stream().synchronize()
+
+    def _bagua_broadcast_optimizer_state(self, optimizer, comm):
+        # Horovod-derived: broadcast tensor state entries, plus pickled
+        # scalar entries (reference: bagua_distributed.py:243-313)
+        if len(optimizer.state_dict()["state"]) == 0:
+            for group in optimizer.param_groups:
+                for p in group["params"]:
+                    if p.requires_grad and id(p) not in optimizer.state:
+                        p.grad = p.data.new(p.size()).zero_()
+            try:
+                optimizer.step()
+                optimizer.zero_grad()
+            except Exception:
+                logger.debug(
+                    "cannot materialize optimizer state for broadcast; "
+                    "skipping (will sync after first step)")
+
+        state = optimizer.state_dict()["state"]
+        tensors = []
+        scalars = {}
+        for pid, pstate in sorted(state.items()):
+            for key, value in sorted(pstate.items()):
+                if isinstance(value, torch.Tensor):
+                    tensors.append(value.data)
+                else:
+                    scalars["{}_{}".format(pid, key)] = value
+        if tensors:
+            broadcast_coalesced(tensors, src=0, comm=comm)
+        if scalars:
+            synced = broadcast_object(scalars, src=0, comm=comm)
+            for pid, pstate in state.items():
+                for key in list(pstate.keys()):
+                    k = "{}_{}".format(pid, key)
+                    if not isinstance(pstate[key], torch.Tensor) \
+                            and k in synced:
+                        pstate[key] = synced[k]
+
+    # ------------------------------------------------------------------
+    # forward-pre hooks (reference: bagua_distributed.py:93-148)
+    # ------------------------------------------------------------------
+    def _install_forward_pre_hooks(self):
+        def num_iteration_step_hook(module, input):
+            if module.training:
+                self.bagua_train_step_counter += 1
+
+        def algorithm_reset_hook(module, input):
+            if module.training and self.bagua_algorithm.need_reset():
+                self._bagua_init_algorithm()
+
+        def algorithm_forward_pre_hook(module, input):
+            if module.training:
+                self.bagua_algorithm.init_forward_pre_hook(self)(input)
+
+        def record_speed_metrics_event(module, input):
+            if module.training:
+                self._speed_metrics_start = time.time()
+
+        def autotune_hook(module, input):
+            if module.training and not self._autotune_completed \
+                    and env.get_autotune_level() >= 1:
+                self._bagua_autotune_step()
+
+        def clear_post_backward_callback_queued_hook(module, input):
+            if module.training:
+                self._is_post_backward_callback_queued = False
+
+        self.module.register_forward_pre_hook(num_iteration_step_hook)
+        self.module.register_forward_pre_hook(autotune_hook)
+        self.module.register_forward_pre_hook(algorithm_reset_hook)
+        self.module.register_forward_pre_hook(algorithm_forward_pre_hook)
+        self.module.register_forward_pre_hook(record_speed_metrics_event)
+        self.module.register_forward_pre_hook(
+            clear_post_backward_callback_queued_hook)
+        self._is_post_backward_callback_queued = False
+
+    # ------------------------------------------------------------------
+    # algorithm (re)initialization (reference: bagua_distributed.py:393-404)
+    # ------------------------------------------------------------------
+    def _bagua_init_algorithm(self):
+        self._cleanup_autograd_hooks()
+        self._bagua_broadcast_parameters()
+        self.tensors = self.bagua_algorithm.init_tensors(self)
+        self._bagua_autotune_register_tensors()
+        self._reset_buckets()
+        self._register_autograd_hooks()
+        self._register_optimizer_hooks()
+
+    def _bucket_suggestion(self) -> List[List[BaguaTensor]]:
+        if self._autotune_client is not None:
+            hp = self._ask_hyperparameters()
+            if hp is not None and hp.buckets:
+                by_name = {t.name: t for t in self.tensors}
+                groups = []
+                for bucket_decl in hp.buckets:
+                    group = [by_name[d.name] for d in bucket_decl
+                             if d.name in by_name]
+                    if group:
+                        groups.append(group)
+                covered = {t.name for g in groups for t in g}
+                rest = [t for t in self.tensors if t.name not in covered]
+                if rest:
+                    groups.extend(split_tensors_into_groups(
+                        rest, env.get_default_bucket_size()))
+                return groups
+        return split_tensors_into_groups(
+            self.tensors, env.get_default_bucket_size())
+
+    def _reset_buckets(self):
+        self.bagua_backend.wait_pending_comm_ops_host()
+        groups = self._bucket_suggestion()
+        self.bagua_buckets = self.bagua_algorithm.tensors_to_buckets(
+            groups, self.gradient_as_bucket_view)
+        for bucket in self.bagua_buckets:
+            self.bagua_algorithm.init_operations(self, bucket)
+        self.bagua_backend.register_ordered_buckets(self.bagua_buckets)
+
+    # ------------------------------------------------------------------
+    # autograd + optimizer hooks (reference: bagua_distributed.py:417-481)
+    # ------------------------------------------------------------------
+    def _cleanup_autograd_hooks(self):
+        for h in self._bagua_autograd_hook_handles:
+            h.remove()
+        self._bagua_autograd_hook_handles = []
+
+    def _register_autograd_hooks(self):
+        backward_hook = self.bagua_algorithm.init_backward_hook(self)
+
+        def make_hook(name):
+            def hook(param):
+                if not self.require_backward_grad_sync:
+                    return
+                bt = self._bagua_tensor_map.get(name)
+                if bt is not None:
+                    bt.repair_bucket_view()
+                backward_hook(name, param)
+                if not self._is_post_backward_callback_queued:
+                    Variable._execution_engine.queue_callback(
+                        self._real_post_backward_hook)
+                    self._is_post_backward_callback_queued = True
+
+            return hook
+
+        for name, param in self.bagua_build_params():
+            handle = param.register_post_accumulate_grad_hook(
+                make_hook(name))
+            self._bagua_autograd_hook_handles.append(handle)
+
+    def _real_post_backward_hook(self):
+        self.bagua_algorithm.init_post_backward_hook(self)()
+        if self._speed_metrics_start is not None:
+            elapsed = time.time() - self._speed_metrics_start
+            total_bytes = sum(b.bytes for b in self.bagua_buckets)
+            if elapsed > 0:
+                self._speed_samples.append(total_bytes / elapsed)
+                del self._speed_samples[:-64]
+            self._speed_metrics_start = None
+
+    def _register_optimizer_hooks(self):
+        post_step = self.bagua_algorithm.init_post_optimizer_step_hook(self)
+        for optimizer in self.bagua_optimizers:
+            if not hasattr(optimizer, "_bagua_original_step"):
+                optimizer._bagua_original_step = optimizer.step
+
+            def make_step(opt):
+                def step(closure=None):
+                    result = opt._bagua_original_step(closure)
+                    post_step(opt)
+                    return result
+
+                return step
+
+            optimizer.step = make_step(optimizer)
+
+    # ------------------------------------------------------------------
+    # autotune client (reference: bagua_distributed.py:325-391)
+    # ------------------------------------------------------------------
+    def _tensor_declarations(self) -> List[TensorDeclaration]:
+        decls = []
+        for t in self.tensors:
+            eff = t.tensor()
+            decls.append(TensorDeclaration(
+                name=t.name, num_elements=eff.numel(),
+                dtype=_TORCH_DTYPE_MAP[eff.dtype]))
+        return decls
+
+    def _bagua_autotune_register_tensors(self):
+        if env.get_autotune_level() < 1:
+            return
+        from ..communication import get_hyperparameters_service_client
+
+        if self._autotune_client is None:
+            self._autotune_client = get_hyperparameters_service_client()
+        self._autotune_client.register_tensors(
+            model_name=self.bagua_module_name,
+            tensor_list=[d.dict() for d in self._tensor_declarations()])
+
+    def _ask_hyperparameters(self) -> Optional[BaguaHyperparameter]:
+        rsp = self._autotune_client.ask_hyperparameters(
+            model_name=self.bagua_module_name,
+            rank=env.get_rank(),
+            train_iter=self.bagua_train_step_counter)
+        if rsp is None:
+            return None
+        hp = BaguaHyperparameter()
+        hp.update(rsp.get("recommended_hyperparameters", {}))
+        self._autotune_completed = bool(rsp.get("is_autotune_completed",
+                                                False))
+        return hp
+
+    def _bagua_autotune_step(self):
+        CYCLE = 100
+        if self.bagua_train_step_counter % CYCLE != 1 \
+                or self.bagua_train_step_counter <= 1:
+            return
+        speed = (sum(self._speed_samples) / len(self._speed_samples)
+                 if self._speed_samples else 0.0)
+        self._autotune_client.report_metrics(
+            model_name=self.bagua_module_name,
+            rank=env.get_rank(),
+            train_iter=self.bagua_train_step_counter,
+            hyperparameters=BaguaHyperparameter(
+                bucket_size=env.get_default_bucket_size()).dict(),
+            speed=speed)
+        self._reset_buckets()
+
+    # ------------------------------------------------------------------
+    def forward(self, *inputs, **kwargs):
+        return self.module(*inputs, **kwargs)
+
+    def __call__(self, *inputs, **kwargs):
+        return self.forward(*inputs, **kwargs)

@@ -1,0 +1,74 @@
+"""Spawn-N-process single-node test harness.
+
+Mirrors the reference's test pattern (tests/internal/multi_process.py:9-53):
+each worker gets a hand-rolled env (WORLD_SIZE/RANK/LOCAL_RANK/MASTER_*)
+on a fresh port, runs the target function, and ships results back through
+a multiprocessing queue. On the CPU CI this exercises the gloo path; the
+same harness with cuda devices runs under ``-m gpu`` on the MI355X box.
+"""
+
+import multiprocessing as mp
+import os
+import pickle
+import socket
+import traceback
+
+
+def find_free_port() -> int:
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _worker(fn, rank, nprocs, port, args, queue):
+    os.environ["WORLD_SIZE"] = str(nprocs)
+    os.environ["LOCAL_WORLD_SIZE"] = str(nprocs)
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["NODE_RANK"] = "0"
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        result = fn(rank, nprocs, *args)
+        queue.put((rank, "ok", pickle.dumps(result)))
+    except Exception:
+        queue.put((rank, "error", traceback.format_exc()))
+
+
+def run_multi_process(nprocs, fn, args=(), timeout=180):
+    """Run ``fn(rank, nprocs, *args)`` in ``nprocs`` spawned processes.
+    Returns results ordered by rank. Raises on any worker failure."""
+    ctx = mp.get_context("spawn")
+    queue = ctx.Queue()
+    port = find_free_port()
+    procs = []
+    for rank in range(nprocs):
+        p = ctx.Process(target=_worker,
+                        args=(fn, rank, nprocs, port, args, queue))
+        p.start()
+        procs.append(p)
+
+    results = {}
+    errors = []
+    for _ in range(nprocs):
+        try:
+            rank, status, payload = queue.get(timeout=timeout)
+        except Exception:
+            for p in procs:
+                p.terminate()
+            raise TimeoutError("worker result timeout after %ss" % timeout)
+        if status == "ok":
+            results[rank] = pickle.loads(payload)
+        else:
+            errors.append("rank %d:\n%s" % (rank, payload))
+
+    for p in procs:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+
+    if errors:
+        raise RuntimeError("worker failures:\n" + "\n".join(errors))
+    return [results[r] for r in range(nprocs)]
