@@ -1,0 +1,187 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — VGG16 synthetic-ImageNet data-parallel training.
+
+Measures the BASELINE.json metric: images/sec (whole node) for VGG16 with
+GradientAllReduce (or ByteGrad via --algorithm) at 1/2/4/8 MI355X.
+
+Single GPU:   python bench.py --gpus 1 --steps 30 --warmup 10
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Timing: W untimed warmup steps, then exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides; elapsed is MAX over ranks;
+rank 0 prints one JSON line.
+
+vs_baseline compares against the reference's strongest per-GPU CI floor on
+its V100 rig: 185.0 img/s/GPU for gradient_allreduce, 180.0 for bytegrad
+(.buildkite/scripts/benchmark_master.sh:81-83), scaled by the
+GPU count of this run.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.nn.functional as F
+
+REFERENCE_PER_GPU_FLOOR = {
+    "gradient_allreduce": 185.0,
+    "bytegrad": 180.0,
+    "decentralized": 150.0,
+    "low_precision_decentralized": 115.0,
+    "async": 190.0,
+    "qadam": 165.0,
+}
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--model", type=str, default="vgg16")
+    p.add_argument("--algorithm", type=str, default="gradient_allreduce")
+    p.add_argument("--batch-size", type=int, default=32,
+                   help="per-GPU batch size")
+    p.add_argument("--dtype", type=str, default="bf16",
+                   choices=["bf16", "fp32"])
+    p.add_argument("--seq-len", type=int, default=384,
+                   help="sequence length for bert-large")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    use_cuda = torch.cuda.is_available()
+
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        torch.backends.cudnn.benchmark = True
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    import bagua_amd
+    from bagua_amd.models import create_model
+    from bagua_amd.parallel.algorithms import GlobalAlgorithmRegistry
+
+    bagua_amd.init_process_group()
+
+    torch.manual_seed(42)
+    model = create_model(args.model).to(device)
+
+    if args.algorithm == "qadam":
+        from bagua_amd.parallel.algorithms.q_adam import QAdamOptimizer
+
+        optimizer = QAdamOptimizer(model.parameters(), lr=1e-4,
+                                   warmup_steps=max(args.warmup, 10))
+        algorithm = GlobalAlgorithmRegistry.get("qadam")(optimizer)
+    else:
+        optimizer = torch.optim.SGD(model.parameters(), lr=0.01,
+                                    momentum=0.9)
+        algorithm = GlobalAlgorithmRegistry.get(args.algorithm)()
+
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer], algorithm=algorithm)
+
+    # synthetic data, resident on device (Horovod-style benchmark; the
+    # reference used the same fixed batch per iteration too)
+    is_bert = args.model.startswith("bert")
+    if is_bert:
+        data = torch.randint(0, 30000, (args.batch_size, args.seq_len),
+                             device=device)
+        target_s = torch.randint(0, args.seq_len, (args.batch_size,),
+                                 device=device)
+        target_e = torch.randint(0, args.seq_len, (args.batch_size,),
+                                 device=device)
+    else:
+        data = torch.randn(args.batch_size, 3, 224, 224, device=device)
+        target = torch.randint(0, 1000, (args.batch_size,), device=device)
+
+    use_bf16 = args.dtype == "bf16"
+    amp_ctx = torch.autocast(device_type=device.type, dtype=torch.bfloat16,
+                             enabled=use_bf16)
+
+    def step():
+        optimizer.zero_grad()
+        with amp_ctx:
+            if is_bert:
+                s_logits, e_logits = ddp(data)
+                loss = (F.cross_entropy(s_logits, target_s)
+                        + F.cross_entropy(e_logits, target_e))
+            else:
+                out = ddp(data)
+                loss = F.cross_entropy(out, target)
+        loss.backward()
+        optimizer.step()
+        return loss
+
+    def barrier_sync():
+        bagua_amd.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+    bagua_amd.allreduce_inplace(t, op=bagua_amd.ReduceOp.MAX)
+    if use_cuda:
+        torch.cuda.synchronize()
+    elapsed_max = float(t.item())
+
+    if args.algorithm == "async":
+        ddp.inner.bagua_algorithm.abort(ddp)
+
+    if rank == 0:
+        unit = "tokens/s" if is_bert else "images/s"
+        per_step_items = args.batch_size * world_size * (
+            args.seq_len if is_bert else 1)
+        value = args.steps * per_step_items / elapsed_max
+        floor = REFERENCE_PER_GPU_FLOOR.get(args.algorithm)
+        vs_baseline = (value / (floor * world_size)
+                       if (floor and not is_bert) else None)
+        result = {
+            "metric": ("images/sec (whole node), VGG16 GradientAllReduce "
+                       "vs ByteGrad at 1/2/4/8 MI355X"
+                       if args.model == "vgg16"
+                       else "%s %s throughput" % (args.model,
+                                                  args.algorithm)),
+            "value": value,
+            "unit": unit,
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed_max / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": vs_baseline,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "algorithm": args.algorithm,
+                "global_batch": args.batch_size * world_size,
+                "seq_len": args.seq_len if is_bert else None,
+                "image_size": None if is_bert else 224,
+                "parallelism": "dp%d" % world_size,
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

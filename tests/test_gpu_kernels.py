@@ -1,0 +1,157 @@
+"""Native CDNA4 kernel numerics vs the pure-torch fp32 references
+(bagua_amd/ops/quant.py and torch ops). GPU-only."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs MI355X")
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float16,
+                                   torch.bfloat16])
+@pytest.mark.parametrize("n", [1024, 4097, 1 << 20])
+def test_elementwise(dtype, n):
+    from bagua_amd import _C
+
+    torch.manual_seed(0)
+    x = torch.randn(n, device="cuda").to(dtype)
+    y = torch.randn(n, device="cuda").to(dtype)
+
+    atol = 1e-6 if dtype == torch.float32 else 2e-2
+
+    a = x.clone()
+    _C.average_inplace(a, y)
+    ref = ((x.float() + y.float()) / 2).to(dtype)
+    assert torch.allclose(a.float(), ref.float(), atol=atol)
+
+    a = x.clone()
+    _C.add_inplace(a, y)
+    assert torch.allclose(a.float(), (x.float() + y.float()).to(dtype).float(),
+                          atol=atol)
+
+    a = x.clone()
+    _C.substract_inplace(a, y)
+    assert torch.allclose(a.float(), (x.float() - y.float()).to(dtype).float(),
+                          atol=atol)
+
+    a = x.clone()
+    _C.addmul_inplace(a, y, 0.37)
+    assert torch.allclose(a.float(),
+                          (x.float() + 0.37 * y.float()).to(dtype).float(),
+                          atol=atol)
+
+    a = x.clone()
+    _C.divide_inplace(a, 3.0)
+    assert torch.allclose(a.float(), (x.float() / 3.0).to(dtype).float(),
+                          atol=atol)
+
+    a = x.clone()
+    r = torch.randn(n, device="cuda").to(dtype)
+    c = torch.randn(n, device="cuda").to(dtype)
+    _C.async_model_average(a, r, c, 4.0)
+    ref = (x.float() + r.float() / 4.0 - c.float()).to(dtype)
+    assert torch.allclose(a.float(), ref.float(), atol=atol * 2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float16,
+                                   torch.bfloat16])
+@pytest.mark.parametrize("num_chunks,chunk", [(4, 1024), (8, 4099)])
+def test_reduce_chunk(dtype, num_chunks, chunk):
+    from bagua_amd import _C
+
+    torch.manual_seed(1)
+    x = torch.randn(num_chunks * chunk, device="cuda").to(dtype)
+    target = 2 % num_chunks
+
+    ref = x.clone()
+    v = ref.view(num_chunks, -1)
+    red = v.float().mean(0).to(dtype)
+    v[target].copy_(red)
+
+    out = x.clone()
+    _C.reduce_chunk_inplace(out, num_chunks, target, True)
+    atol = 1e-5 if dtype == torch.float32 else 2e-2
+    assert torch.allclose(out.float(), ref.float(), atol=atol)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float16,
+                                   torch.bfloat16])
+@pytest.mark.parametrize("num_chunks,chunk", [(1, 4096), (4, 1024),
+                                              (8, 32768)])
+def test_compress_matches_golden(dtype, num_chunks, chunk):
+    """uint8 payload must match the pure-torch golden quantizer bitwise
+    (modulo rint ties at .5, which both sides resolve to even)."""
+    from bagua_amd import _C
+    from bagua_amd.ops import quant
+
+    torch.manual_seed(2)
+    x = (torch.randn(num_chunks * chunk, device="cuda") * 3).to(dtype)
+
+    stride = quant.compressed_chunk_bytes(chunk)
+    out = torch.zeros(stride * num_chunks, dtype=torch.uint8, device="cuda")
+    _C.compress_chunked(x, out, num_chunks, -1)
+    torch.cuda.synchronize()
+
+    gold = quant.compress_chunked(x.float(), num_chunks)
+    # headers: min/max stored as f32 from the input dtype's values
+    for c in range(num_chunks):
+        hdr = out[c * stride:c * stride + 8].view(torch.float32)
+        ghdr = gold[c * stride:c * stride + 8].view(torch.float32)
+        assert torch.allclose(hdr, ghdr, atol=1e-6), "chunk %d header" % c
+        payload = out[c * stride + 32:c * stride + 32 + chunk]
+        gpayload = gold[c * stride + 32:c * stride + 32 + chunk]
+        diff = (payload.int() - gpayload.int()).abs()
+        # float->dtype->float rounding can move rint by 1 level for
+        # f16/bf16; f32 must be exact
+        if dtype == torch.float32:
+            assert int(diff.max()) == 0
+        else:
+            assert int(diff.max()) <= 1
+
+    # roundtrip through native decompress
+    y = torch.zeros_like(x)
+    _C.decompress_chunked(out, y, num_chunks, -1)
+    torch.cuda.synchronize()
+    v = x.float().view(num_chunks, -1)
+    step = ((v.max(1).values - v.min(1).values) / 255.0).max()
+    err = (x.float() - y.float()).abs().max()
+    assert err <= step + step * 0.5 + 1e-5
+
+
+@requires_gpu
+def test_single_rank_communicator():
+    """1-GPU RCCL communicator: allreduce/broadcast/allgather are
+    identity; exercises unique-id exchange and stream discipline."""
+    import os
+
+    os.environ.setdefault("WORLD_SIZE", "1")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("LOCAL_RANK", "0")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29581")
+
+    import bagua_amd
+    from bagua_amd.communication import ReduceOp
+
+    torch.cuda.set_device(0)
+    bagua_amd.init_process_group()
+    comm = bagua_amd.communication._get_default_group() \
+        .get_global_communicator()
+    comm.ensure_native()
+    assert comm.is_native, "native RCCL communicator must be active on GPU"
+
+    x = torch.randn(1024, device="cuda")
+    ref = x.clone()
+    comm.allreduce_inplace(x, ReduceOp.SUM)
+    torch.cuda.synchronize()
+    assert torch.allclose(x, ref)
+
+    comm.broadcast(x, 0)
+    torch.cuda.synchronize()
+    assert torch.allclose(x, ref)
