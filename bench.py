@@ -49,6 +49,9 @@ def parse_args():
                    choices=["bf16", "fp32"])
     p.add_argument("--seq-len", type=int, default=384,
                    help="sequence length for bert-large")
+    p.add_argument("--no-channels-last", action="store_true",
+                   help="disable NHWC layout for conv models (NHWC is the "
+                        "MIOpen fast path on MI355X)")
     return p.parse_args()
 
 
@@ -74,6 +77,11 @@ def main():
 
     torch.manual_seed(42)
     model = create_model(args.model).to(device)
+    is_conv_model = not args.model.startswith("bert")
+    channels_last = (is_conv_model and use_cuda
+                     and not args.no_channels_last)
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
 
     if args.algorithm == "qadam":
         from bagua_amd.parallel.algorithms.q_adam import QAdamOptimizer
@@ -101,6 +109,8 @@ def main():
                                  device=device)
     else:
         data = torch.randn(args.batch_size, 3, 224, 224, device=device)
+        if channels_last:
+            data = data.to(memory_format=torch.channels_last)
         target = torch.randint(0, 1000, (args.batch_size,), device=device)
 
     use_bf16 = args.dtype == "bf16"
