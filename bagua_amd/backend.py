@@ -18,10 +18,12 @@ Re-design of the reference's Rust BaguaCommBackend
 """
 
 import logging
+import time
 from typing import List, Optional
 
 import torch
 
+from . import env
 from .bucket import BaguaBucket
 from .executor import execute_ops
 from .tensor import BaguaTensor
@@ -52,6 +54,10 @@ class BaguaBackend:
         self._queue_idx = 0
         self._done_events: List[torch.cuda.Event] = []
         self._tensor_names = set()
+        # telemetry: tensor-ready spans for the autotune execution-order
+        # pipeline (reference: lib.rs:305-307 + bagua-opentelemetry)
+        self.telemetry_enabled = env.get_autotune_level() > 0
+        self.telemetry_spans: List[dict] = []
 
     # ------------------------------------------------------------------
     def register_ordered_buckets(self, buckets: List[BaguaBucket]):
@@ -83,6 +89,12 @@ class BaguaBackend:
         """Set the tensor ready; launch every fully-ready front bucket
         in order (reference: lib.rs:300-319)."""
         btensor.ready = True
+        if self.telemetry_enabled and len(self.telemetry_spans) < 10000:
+            now = time.monotonic_ns()
+            self.telemetry_spans.append({
+                "trace_id": 0, "action": "tensor_ready",
+                "tensor_name": btensor.name,
+                "start_time": now, "end_time": now})
         while self._queue_idx < len(self.ordered_buckets):
             bucket = self.ordered_buckets[self._queue_idx]
             if not bucket.ready_for_comm():

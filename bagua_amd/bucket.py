@@ -124,7 +124,7 @@ class BaguaBucket:
         return sum(t.numel() * t.tensor().element_size() for t in self.tensors)
 
     def _flatten_(self, alignment: int):
-        effs = [t.tensor() for t in self.tensors]
+        effs = [t.materialized() for t in self.tensors]
         total = sum(e.numel() for e in effs)
         if alignment > 1:
             total = ((total + alignment - 1) // alignment) * alignment

@@ -16,6 +16,7 @@ PyTorch on ROCm:
 """
 
 import logging
+import os
 import time
 from typing import Callable, Dict, List, Optional
 
@@ -108,6 +109,8 @@ class BaguaDistributedDataParallel:
 
         self._autotune_client = None
         self._autotune_completed = env.get_autotune_level() == 0
+        self._current_hp = BaguaHyperparameter(
+            bucket_size=env.get_default_bucket_size())
         self._bagua_autograd_hook_handles = []
         self._hooks_fired_this_iter = False
 
@@ -367,14 +370,14 @@ class BaguaDistributedDataParallel:
             train_iter=self.bagua_train_step_counter)
         if rsp is None:
             return None
-        hp = BaguaHyperparameter()
-        hp.update(rsp.get("recommended_hyperparameters", {}))
+        hp = BaguaHyperparameter(**rsp.get("recommended_hyperparameters", {}))
         self._autotune_completed = bool(rsp.get("is_autotune_completed",
                                                 False))
+        self._current_hp = hp
         return hp
 
     def _bagua_autotune_step(self):
-        CYCLE = 100
+        CYCLE = int(os.environ.get("BAGUA_AUTOTUNE_INTERVAL", 100))
         if self.bagua_train_step_counter % CYCLE != 1 \
                 or self.bagua_train_step_counter <= 1:
             return
@@ -384,9 +387,14 @@ class BaguaDistributedDataParallel:
             model_name=self.bagua_module_name,
             rank=env.get_rank(),
             train_iter=self.bagua_train_step_counter,
-            hyperparameters=BaguaHyperparameter(
-                bucket_size=env.get_default_bucket_size()).dict(),
+            hyperparameters=self._current_hp.dict(),
             speed=speed)
+        spans = self.bagua_backend.telemetry_spans
+        if spans:
+            for s in spans:
+                s["model_name"] = self.bagua_module_name
+            self._autotune_client.report_tensor_execution_order(spans[-512:])
+            self.bagua_backend.telemetry_spans = []
         self._reset_buckets()
 
     # ------------------------------------------------------------------

@@ -54,6 +54,18 @@ class BaguaTensor:
             return self.getter_closure(self.proxy)
         return self.proxy
 
+    def materialized(self) -> torch.Tensor:
+        """The effective tensor, materializing it (zeros shaped like the
+        proxy) when a grad-getter currently returns None — happens when a
+        re-bucket lands between zero_grad(set_to_none) and backward."""
+        eff = self.tensor()
+        if eff is None:
+            fresh = torch.zeros_like(self.proxy)
+            if self.setter_closure is not None:
+                self.setter_closure(self.proxy, fresh)
+            eff = self.tensor()
+        return eff
+
     def data_ptr(self) -> int:
         return self.tensor().data_ptr()
 
