@@ -18,6 +18,7 @@ Re-design of the reference's Rust BaguaCommBackend
 """
 
 import logging
+import os
 import time
 from typing import List, Optional
 
@@ -58,6 +59,7 @@ class BaguaBackend:
         # pipeline (reference: lib.rs:305-307 + bagua-opentelemetry)
         self.telemetry_enabled = env.get_autotune_level() > 0
         self.telemetry_spans: List[dict] = []
+        self._native_exec = None  # C++ BucketExecutor (GPU)
 
     # ------------------------------------------------------------------
     def register_ordered_buckets(self, buckets: List[BaguaBucket]):
@@ -66,6 +68,7 @@ class BaguaBackend:
         self.wait_pending_comm_ops()
         if self.group.stream is not None:
             self.group.stream.synchronize()
+        self._register_native(buckets)
         # duplicate detection (reference: lib.rs:282-295)
         names = set()
         ptrs = set()
@@ -83,6 +86,61 @@ class BaguaBackend:
         self.ordered_buckets = list(buckets)
         self._tensor_names = names
         self._queue_idx = 0
+
+    # ------------------------------------------------------------------
+    # native C++ bucket executor (GPU hot path; see csrc/core.cpp
+    # BucketExecutor). Eligible buckets — flattened, single centralized
+    # op, supported dtype — run entirely in C++ with the GIL released.
+    # ------------------------------------------------------------------
+    def _register_native(self, buckets: List[BaguaBucket]):
+        from .bucket import CentralizedSyncOp
+        from .ops import native
+
+        for b in buckets:
+            b._native_idx = None
+        if not (torch.cuda.is_available() and native.available()
+                and self.group.stream is not None):
+            return
+        if os.environ.get("BAGUA_NATIVE_SCHEDULER", "1") != "1":
+            return
+
+        def eligible(b: BaguaBucket) -> bool:
+            if b._flat is None or len(b.ops) != 1:
+                return False
+            op = b.ops[0]
+            if not isinstance(op, CentralizedSyncOp):
+                return False
+            if op.compression not in (None, "MinMaxUInt8"):
+                return False
+            return b._flat.dtype in (torch.float32, torch.float16,
+                                     torch.bfloat16)
+
+        todo = [b for b in buckets if eligible(b)]
+        if not todo:
+            return
+        if self._native_exec is None:
+            glob = self.group.get_global_communicator()
+            glob.ensure_native()
+            intra = inter = None
+            hierarchical = any(b.ops[0].hierarchical for b in todo)
+            if hierarchical:
+                intra_c = self.group.get_intra_node_communicator()
+                if intra_c.nranks() < glob.nranks():
+                    intra_c.ensure_native()
+                    intra = intra_c._native
+                    if intra_c.rank_in_comm == 0:
+                        inter_c = self.group.get_inter_node_communicator()
+                        inter_c.ensure_native()
+                        inter = inter_c._native
+            self._native_exec = native.lib().BucketExecutor(
+                glob._native, intra, inter, self.group.stream.cuda_stream)
+        else:
+            self._native_exec.clear_buckets()
+        for b in todo:
+            op = b.ops[0]
+            b._native_idx = self._native_exec.register_bucket(
+                b._flat, op.compression == "MinMaxUInt8",
+                op.scattergather, op.average, op.hierarchical)
 
     # ------------------------------------------------------------------
     def mark_communication_ready(self, btensor: BaguaTensor):
@@ -106,6 +164,17 @@ class BaguaBackend:
             self._queue_idx = 0
 
     def _execute(self, bucket: BaguaBucket):
+        if getattr(bucket, "_native_idx", None) is not None:
+            events = []
+            for t in bucket.tensors:
+                if t.ready_event is not None:
+                    events.append(t.ready_event.cuda_event)
+            self._native_exec.execute(bucket._native_idx, events)
+            for t in bucket.tensors:
+                if t.ready_event is not None:
+                    self.event_pool.put(t.ready_event)
+                    t.ready_event = None
+            return
         if self.group.stream is not None and torch.cuda.is_available():
             comm_stream = self.group.stream
             for t in bucket.tensors:
@@ -136,11 +205,16 @@ class BaguaBackend:
                 curr.wait_event(ev)
                 self.event_pool.put(ev)
             self._done_events.clear()
+        if self._native_exec is not None:
+            self._native_exec.wait_pending(
+                torch.cuda.current_stream().cuda_stream)
         return n
 
     def wait_pending_comm_ops_host(self) -> int:
         """Host-blocking variant (used before rebucketing / shutdown)."""
         n = self.wait_pending_comm_ops()
+        if self._native_exec is not None:
+            self._native_exec.synchronize()
         if self.group.stream is not None:
             self.group.stream.synchronize()
         return n

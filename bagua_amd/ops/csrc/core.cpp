@@ -15,9 +15,15 @@
 
 #include <c10/hip/HIPStream.h>
 
+#include <chrono>
+#include <condition_variable>
 #include <cstring>
+#include <memory>
+#include <mutex>
 #include <stdexcept>
 #include <string>
+#include <thread>
+#include <vector>
 
 // kernels.hip launchers
 extern "C" {
@@ -138,6 +144,8 @@ class Communicator {
       comm_ = nullptr;
     }
   }
+
+  ncclComm_t raw() const { return comm_; }
 
   void group_start() { NCCL_CHECK(ncclGroupStart()); }
   void group_end() { NCCL_CHECK(ncclGroupEnd()); }
@@ -326,6 +334,233 @@ class Communicator {
   ncclComm_t comm_ = nullptr;
 };
 
+static int64_t compressed_chunk_stride(int64_t chunk);
+
+// ---------------------------------------------------------------------------
+// BucketExecutor — native centralized-op executor + watchdog.
+//
+// MI355X re-design of the reference's background scheduler
+// (bagua-core-internal/src/lib.rs:209-338): instead of a comm worker
+// thread draining a channel, the autograd hook thread calls execute()
+// directly — RCCL calls are enqueue-only, so a thread handoff would only
+// add latency. The GIL is released for the whole bucket execution, all
+// work lands on the dedicated comm stream, and completion is tracked by
+// events the compute stream waits on (no host syncs in the hot path).
+// The watchdog thread reproduces comm_monitor (lib.rs:255-265): if a
+// scheduled bucket has not completed within 300 s it aborts the
+// communicator and screams, so a wedged collective cannot hang training
+// silently.
+// ---------------------------------------------------------------------------
+
+class BucketExecutor {
+ public:
+  struct Bucket {
+    at::Tensor flat;
+    bool compressed = false;
+    bool scattergather = false;
+    bool average = true;
+    bool hierarchical = false;
+    at::Tensor wire;       // compressed wire buffer
+    at::Tensor wire_tmp;   // alltoall bounce
+    at::Tensor scratch;    // minmax scratch (uint32 x 2*chunks)
+  };
+
+  BucketExecutor(std::shared_ptr<Communicator> global,
+                 std::shared_ptr<Communicator> intra,
+                 std::shared_ptr<Communicator> inter, uintptr_t stream)
+      : global_(std::move(global)), intra_(std::move(intra)),
+        inter_(std::move(inter)), stream_((hipStream_t)stream),
+        stop_(false) {
+    watchdog_ = std::thread([this] { this->watch(); });
+  }
+
+  ~BucketExecutor() {
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      stop_ = true;
+    }
+    cv_.notify_all();
+    if (watchdog_.joinable()) watchdog_.join();
+    for (auto ev : event_pool_) hipEventDestroy(ev);
+    for (auto& p : inflight_) hipEventDestroy(p.first);
+  }
+
+  int register_bucket(at::Tensor flat, bool compressed, bool scattergather,
+                      bool average, bool hierarchical) {
+    check_device_contig(flat);
+    Bucket b;
+    b.flat = flat;
+    b.compressed = compressed;
+    b.scattergather = scattergather;
+    b.average = average;
+    b.hierarchical = hierarchical;
+    if (compressed) {
+      Communicator* c = inner_comm(hierarchical);
+      if (c != nullptr) {  // null on hierarchical non-leader ranks
+        int n = c->nranks();
+        TORCH_CHECK(flat.numel() % n == 0, "bucket not padded to nranks");
+        int64_t chunk = flat.numel() / n;
+        int64_t stride = compressed_chunk_stride(chunk);
+        b.wire = at::empty({stride * n}, flat.options().dtype(at::kByte));
+        b.wire_tmp = at::empty_like(b.wire);
+        b.scratch = at::empty({2 * n}, flat.options().dtype(at::kInt));
+      }
+    }
+    buckets_.push_back(std::move(b));
+    return (int)buckets_.size() - 1;
+  }
+
+  void execute(int idx, const std::vector<uintptr_t>& ready_events) {
+    py::gil_scoped_release nogil;
+    Bucket& b = buckets_.at(idx);
+    for (uintptr_t ev : ready_events)
+      HIP_CHECK(hipStreamWaitEvent(stream_, (hipEvent_t)ev, 0));
+
+    bool multi_node = b.hierarchical && intra_ &&
+                      intra_->nranks() < global_->nranks();
+    Communicator* comm = multi_node ? inter_.get() : global_.get();
+
+    if (multi_node)
+      intra_->reduce_inplace(b.flat, 0, b.average ? 10 : 0);
+    if (!multi_node || intra_->rank() == 0) {
+      TORCH_CHECK(comm != nullptr, "leader rank missing inter communicator");
+      if (b.compressed)
+        run_compressed(b, comm);
+      else if (b.scattergather)
+        run_scattergather(b, comm);
+      else
+        comm->allreduce_inplace(b.flat, b.average ? 10 : 0);
+    }
+    if (multi_node) intra_->broadcast(b.flat, 0);
+
+    hipEvent_t done = get_event();
+    HIP_CHECK(hipEventRecord(done, stream_));
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      inflight_.emplace_back(done, std::chrono::steady_clock::now());
+    }
+  }
+
+  // make the given (compute) stream wait on all scheduled comm; recycles
+  // completed events. GPU-side only — never blocks the host.
+  void wait_pending(uintptr_t compute_stream) {
+    py::gil_scoped_release nogil;
+    std::lock_guard<std::mutex> g(mu_);
+    for (auto& p : inflight_) {
+      HIP_CHECK(hipStreamWaitEvent((hipStream_t)compute_stream, p.first,
+                                   0));
+      event_pool_.push_back(p.first);
+    }
+    inflight_.clear();
+  }
+
+  void synchronize() {
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    std::lock_guard<std::mutex> g(mu_);
+    for (auto& p : inflight_) event_pool_.push_back(p.first);
+    inflight_.clear();
+  }
+
+  void clear_buckets() { buckets_.clear(); }
+
+ private:
+  Communicator* inner_comm(bool hierarchical) {
+    bool multi_node = hierarchical && intra_ &&
+                      intra_->nranks() < global_->nranks();
+    if (!multi_node) return global_.get();
+    return inter_ ? inter_.get() : nullptr;
+  }
+
+  void run_scattergather(Bucket& b, Communicator* comm) {
+    int n = comm->nranks();
+    comm->alltoall_inplace(b.flat);
+    bagua_reduce_chunk_launch(kernel_dtype(b.flat), b.flat.data_ptr(), n,
+                              comm->rank(), b.average ? 1 : 0,
+                              b.flat.numel() / n, stream_);
+    comm->allgather_inplace(b.flat);
+  }
+
+  void run_compressed(Bucket& b, Communicator* comm) {
+    // ByteGrad wire protocol (reference:
+    // comm_ops/centralized_low_precision_synchronous.rs:16-74)
+    int n = comm->nranks();
+    int rank = comm->rank();
+    int64_t chunk = b.flat.numel() / n;
+    int64_t stride = compressed_chunk_stride(chunk);
+    int dt = kernel_dtype(b.flat);
+    auto* wire = (uint8_t*)b.wire.data_ptr();
+    auto* scratch = (uint32_t*)b.scratch.data_ptr();
+
+    bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, chunk,
+                          stride, n, 0, n, stream_);
+    comm->alltoall(b.wire, b.wire_tmp);
+    std::swap(b.wire, b.wire_tmp);
+    wire = (uint8_t*)b.wire.data_ptr();
+    bagua_decompress_launch(dt, wire, b.flat.data_ptr(), chunk, stride, 0,
+                            n, stream_);
+    bagua_reduce_chunk_launch(dt, b.flat.data_ptr(), n, rank,
+                              b.average ? 1 : 0, chunk, stream_);
+    bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, chunk,
+                          stride, n, rank, 1, stream_);
+    // in-place allgather of the rank's wire chunk
+    {
+      char* base = (char*)wire;
+      char* own = base + (int64_t)rank * stride;
+      NCCL_CHECK(ncclAllGather(own, base, stride, ncclUint8,
+                               comm->raw(), stream_));
+    }
+    bagua_decompress_launch(dt, wire, b.flat.data_ptr(), chunk, stride, 0,
+                            n, stream_);
+  }
+
+  hipEvent_t get_event() {
+    std::lock_guard<std::mutex> g(mu_);
+    if (!event_pool_.empty()) {
+      hipEvent_t ev = event_pool_.back();
+      event_pool_.pop_back();
+      return ev;
+    }
+    hipEvent_t ev;
+    HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    return ev;
+  }
+
+  void watch() {
+    using namespace std::chrono;
+    std::unique_lock<std::mutex> lk(mu_);
+    while (!stop_) {
+      cv_.wait_for(lk, seconds(10));
+      if (stop_) break;
+      auto now = steady_clock::now();
+      for (auto& p : inflight_) {
+        if (hipEventQuery(p.first) == hipSuccess) continue;
+        if (duration_cast<seconds>(now - p.second).count() > 300) {
+          fprintf(stderr,
+                  "[bagua_amd] FATAL: a scheduled communication has been "
+                  "running for >300s; aborting communicator\n");
+          fflush(stderr);
+          lk.unlock();
+          global_->abort();
+          lk.lock();
+          break;
+        }
+      }
+    }
+  }
+
+  std::shared_ptr<Communicator> global_, intra_, inter_;
+  hipStream_t stream_;
+  std::vector<Bucket> buckets_;
+  std::vector<hipEvent_t> event_pool_;
+  std::vector<std::pair<hipEvent_t,
+                        std::chrono::steady_clock::time_point>> inflight_;
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::thread watchdog_;
+  bool stop_;
+};
+
 // ---------------------------------------------------------------------------
 // kernel wrappers (run on the CALLER's current torch stream so they are
 // ordered with collectives when invoked under torch.cuda.stream(comm))
@@ -418,7 +653,7 @@ static py::bytes nccl_unique_id() {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "bagua_amd native core (RCCL over xGMI + CDNA4 kernels)";
 
-  py::class_<Communicator>(m, "Communicator")
+  py::class_<Communicator, std::shared_ptr<Communicator>>(m, "Communicator")
       .def(py::init<int, int, int, uintptr_t, const std::string&>())
       .def("rank", &Communicator::rank)
       .def("nranks", &Communicator::nranks)
@@ -442,6 +677,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("gather_inplace", &Communicator::gather_inplace)
       .def("scatter", &Communicator::scatter)
       .def("scatter_inplace", &Communicator::scatter_inplace);
+
+  py::class_<BucketExecutor>(m, "BucketExecutor")
+      .def(py::init<std::shared_ptr<Communicator>,
+                    std::shared_ptr<Communicator>,
+                    std::shared_ptr<Communicator>, uintptr_t>(),
+           py::arg("global_comm"), py::arg("intra_comm") = nullptr,
+           py::arg("inter_comm") = nullptr, py::arg("stream") = 0)
+      .def("register_bucket", &BucketExecutor::register_bucket)
+      .def("execute", &BucketExecutor::execute)
+      .def("wait_pending", &BucketExecutor::wait_pending)
+      .def("synchronize", &BucketExecutor::synchronize)
+      .def("clear_buckets", &BucketExecutor::clear_buckets);
 
   m.def("nccl_unique_id", &nccl_unique_id);
   m.def("average_inplace", &average_inplace);

@@ -46,3 +46,82 @@ def test_train_steps(algo):
         optimizer.step()
         losses.append(loss.item())
     assert all(torch.isfinite(torch.tensor(losses)))
+
+
+@requires_gpu
+def test_native_bucket_executor_engaged():
+    """The C++ BucketExecutor must carry eligible buckets on GPU."""
+    _setup_env()
+    import bagua_amd
+    from bagua_amd.models import MnistNet
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+
+    torch.cuda.set_device(0)
+    bagua_amd.init_process_group()
+    torch.manual_seed(4)
+    model = MnistNet().cuda()
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.01)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm())
+    backend = ddp.inner.bagua_backend
+    assert backend._native_exec is not None, "native executor not built"
+    assert any(getattr(b, "_native_idx", None) is not None
+               for b in ddp.inner.bagua_buckets), "no native buckets"
+    data = torch.randn(8, 1, 28, 28, device="cuda")
+    target = torch.randint(0, 10, (8,), device="cuda")
+    before = torch.cat([p.grad.reshape(-1).clone()
+                        for p in model.parameters()
+                        if p.grad is not None]) \
+        if any(p.grad is not None for p in model.parameters()) else None
+    optimizer.zero_grad()
+    loss = F.nll_loss(ddp(data), target)
+    loss.backward()
+    optimizer.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss).item()
+
+
+@requires_gpu
+@pytest.mark.parametrize("algo,kwargs", [
+    ("decentralized", {"peer_selection_mode": "all"}),
+    ("low_precision_decentralized", {}),
+    ("qadam", {}),
+    ("async", {"sync_interval_ms": 10, "warmup_steps": 2}),
+])
+def test_train_steps_other_algorithms(algo, kwargs):
+    _setup_env()
+    import bagua_amd
+    from bagua_amd.models import MnistNet
+    from bagua_amd.parallel.algorithms import GlobalAlgorithmRegistry
+
+    torch.cuda.set_device(0)
+    bagua_amd.init_process_group()
+    torch.manual_seed(3)
+    model = MnistNet().cuda()
+    if algo == "qadam":
+        from bagua_amd.parallel.algorithms.q_adam import QAdamOptimizer
+
+        optimizer = QAdamOptimizer(model.parameters(), lr=1e-3,
+                                   warmup_steps=4)
+        algorithm = GlobalAlgorithmRegistry.get(algo)(optimizer)
+    else:
+        optimizer = torch.optim.SGD(model.parameters(), lr=0.01)
+        algorithm = GlobalAlgorithmRegistry.get(algo)(**dict(kwargs))
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer], algorithm=algorithm)
+    losses = []
+    for step in range(8):
+        data = torch.randn(16, 1, 28, 28, device="cuda")
+        target = torch.randint(0, 10, (16,), device="cuda")
+        optimizer.zero_grad()
+        loss = F.nll_loss(ddp(data), target)
+        loss.backward()
+        optimizer.step()
+        losses.append(loss.item())
+    if algo == "async":
+        ddp.inner.bagua_algorithm.abort(ddp)
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(torch.tensor(losses)))
