@@ -9,3 +9,4 @@ from .cache_loader import CacheLoader  # noqa: F401
 from .cached_dataset import CachedDataset  # noqa: F401
 from .sync_batchnorm import SyncBatchNorm  # noqa: F401
 from . import utils  # noqa: F401
+from .fused_kernels import FusedAdam, FusedAdamW, FusedSGD  # noqa: F401

@@ -16,6 +16,7 @@
 #include <c10/hip/HIPStream.h>
 
 #include <chrono>
+#include <cmath>
 #include <condition_variable>
 #include <cstring>
 #include <memory>
@@ -44,6 +45,15 @@ void bagua_decompress_launch(int dtype, const uint8_t* in, void* x,
                              size_t chunk, size_t chunk_stride,
                              int chunk_begin, int chunk_count,
                              hipStream_t stream);
+void bagua_fused_sgd_launch(float* p, const float* g, float* m, float lr,
+                            float momentum, float dampening,
+                            float weight_decay, int nesterov,
+                            int momentum_initialized, size_t n,
+                            hipStream_t stream);
+void bagua_fused_adam_launch(float* p, const float* g, float* m, float* v,
+                             float lr, float beta1, float beta2, float eps,
+                             float weight_decay, int adamw, float bc1,
+                             float bc2, size_t n, hipStream_t stream);
 }
 
 #define HIP_CHECK(cmd)                                                    \
@@ -644,6 +654,35 @@ static void decompress_chunked(at::Tensor buf, at::Tensor flat,
                           current_stream());
 }
 
+static void fused_sgd_step(at::Tensor p, at::Tensor g, at::Tensor m,
+                           double lr, double momentum, double dampening,
+                           double weight_decay, bool nesterov,
+                           bool momentum_initialized) {
+  check_device_contig(p);
+  TORCH_CHECK(p.scalar_type() == at::kFloat, "fused SGD needs f32 master");
+  bagua_fused_sgd_launch(
+      (float*)p.data_ptr(), (const float*)g.data_ptr(),
+      momentum != 0.0 ? (float*)m.data_ptr() : nullptr, (float)lr,
+      (float)momentum, (float)dampening, (float)weight_decay,
+      nesterov ? 1 : 0, momentum_initialized ? 1 : 0, p.numel(),
+      current_stream());
+}
+
+static void fused_adam_step(at::Tensor p, at::Tensor g, at::Tensor m,
+                            at::Tensor v, int64_t step, double lr,
+                            double beta1, double beta2, double eps,
+                            double weight_decay, bool adamw) {
+  check_device_contig(p);
+  TORCH_CHECK(p.scalar_type() == at::kFloat, "fused Adam needs f32 master");
+  float bc1 = 1.f - (float)std::pow(beta1, (double)step);
+  float bc2 = 1.f - (float)std::pow(beta2, (double)step);
+  bagua_fused_adam_launch(
+      (float*)p.data_ptr(), (const float*)g.data_ptr(),
+      (float*)m.data_ptr(), (float*)v.data_ptr(), (float)lr, (float)beta1,
+      (float)beta2, (float)eps, (float)weight_decay, adamw ? 1 : 0, bc1,
+      bc2, p.numel(), current_stream());
+}
+
 static py::bytes nccl_unique_id() {
   ncclUniqueId id;
   NCCL_CHECK(ncclGetUniqueId(&id));
@@ -701,4 +740,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("compress_chunked", &compress_chunked);
   m.def("decompress_chunked", &decompress_chunked);
   m.def("compressed_chunk_stride", &compressed_chunk_stride);
+  m.def("fused_sgd_step", &fused_sgd_step);
+  m.def("fused_adam_step", &fused_adam_step);
 }

@@ -511,3 +511,115 @@ void bagua_decompress_launch(int dtype, const uint8_t* in, void* x,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// fused optimizer steps (f32 master weights) — one pass over the flat
+// param group: grad transform + momentum/variance update + weight update
+// in a single memory-bound kernel instead of the 3-5 launches of a
+// generic foreach optimizer. MI355X-native addition (the reference fused
+// by contiguity only; north-star asks for the fused step as a HIP
+// kernel).
+// ---------------------------------------------------------------------------
+
+extern "C" {
+
+// SGD with momentum/dampening/nesterov/weight-decay (torch semantics)
+__global__ void fused_sgd_kernel(float* __restrict__ p,
+                                 const float* __restrict__ g,
+                                 float* __restrict__ m, float lr,
+                                 float momentum, float dampening,
+                                 float weight_decay, int nesterov,
+                                 int momentum_initialized, size_t n) {
+  const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const size_t nv = n / 4;
+  float4* p4 = reinterpret_cast<float4*>(p);
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  float4* m4 = reinterpret_cast<float4*>(m);
+  for (size_t i = tid; i < nv; i += stride) {
+    float4 pv = p4[i], gv = g4[i];
+    float4 mv = momentum != 0.f ? m4[i] : make_float4(0, 0, 0, 0);
+    float* pp = &pv.x;
+    const float* gg = &gv.x;
+    float* mm = &mv.x;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float grad = gg[k] + weight_decay * pp[k];
+      float upd = grad;
+      if (momentum != 0.f) {
+        float buf = momentum_initialized
+                        ? momentum * mm[k] + (1.f - dampening) * grad
+                        : grad;
+        mm[k] = buf;
+        upd = nesterov ? grad + momentum * buf : buf;
+      }
+      pp[k] -= lr * upd;
+    }
+    p4[i] = pv;
+    if (momentum != 0.f) m4[i] = mv;
+  }
+  for (size_t i = nv * 4 + tid; i < n; i += stride) {
+    float grad = g[i] + weight_decay * p[i];
+    float upd = grad;
+    if (momentum != 0.f) {
+      float buf = momentum_initialized
+                      ? momentum * m[i] + (1.f - dampening) * grad
+                      : grad;
+      m[i] = buf;
+      upd = nesterov ? grad + momentum * buf : buf;
+    }
+    p[i] -= lr * upd;
+  }
+}
+
+// Adam / AdamW (torch semantics, bias-corrected)
+__global__ void fused_adam_kernel(float* __restrict__ p,
+                                  const float* __restrict__ g,
+                                  float* __restrict__ m,
+                                  float* __restrict__ v, float lr,
+                                  float beta1, float beta2, float eps,
+                                  float weight_decay, int adamw,
+                                  float bias_correction1,
+                                  float bias_correction2, size_t n) {
+  const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const float inv_bc1 = 1.f / bias_correction1;
+  const float inv_sqrt_bc2 = rsqrtf(bias_correction2);
+  for (size_t i = tid; i < n; i += stride) {
+    float param = p[i];
+    float grad = g[i];
+    if (adamw)
+      param *= (1.f - lr * weight_decay);
+    else
+      grad += weight_decay * param;
+    float mi = beta1 * m[i] + (1.f - beta1) * grad;
+    float vi = beta2 * v[i] + (1.f - beta2) * grad * grad;
+    m[i] = mi;
+    v[i] = vi;
+    float denom = sqrtf(vi) * inv_sqrt_bc2 + eps;
+    p[i] = param - lr * inv_bc1 * mi / denom;
+  }
+}
+
+void bagua_fused_sgd_launch(float* p, const float* g, float* m, float lr,
+                            float momentum, float dampening,
+                            float weight_decay, int nesterov,
+                            int momentum_initialized, size_t n,
+                            hipStream_t stream) {
+  int grid = grid_for(n / 4 + 1);
+  hipLaunchKernelGGL(fused_sgd_kernel, dim3(grid), dim3(BLOCK), 0, stream,
+                     p, g, m, lr, momentum, dampening, weight_decay,
+                     nesterov, momentum_initialized, n);
+}
+
+void bagua_fused_adam_launch(float* p, const float* g, float* m, float* v,
+                             float lr, float beta1, float beta2, float eps,
+                             float weight_decay, int adamw, float bc1,
+                             float bc2, size_t n, hipStream_t stream) {
+  int grid = grid_for(n + 1);
+  hipLaunchKernelGGL(fused_adam_kernel, dim3(grid), dim3(BLOCK), 0, stream,
+                     p, g, m, v, lr, beta1, beta2, eps, weight_decay,
+                     adamw, bc1, bc2, n);
+}
+
+}  // extern "C"
