@@ -100,7 +100,11 @@ def main():
     # synthetic data, resident on device (Horovod-style benchmark; the
     # reference used the same fixed batch per iteration too)
     is_bert = args.model.startswith("bert")
-    if is_bert:
+    is_mnist = args.model == "mnist"
+    if is_mnist:  # tiny CPU-testable path for the distributed plumbing
+        data = torch.randn(args.batch_size, 1, 28, 28, device=device)
+        target = torch.randint(0, 10, (args.batch_size,), device=device)
+    elif is_bert:
         data = torch.randint(0, 30000, (args.batch_size, args.seq_len),
                              device=device)
         target_s = torch.randint(0, args.seq_len, (args.batch_size,),
@@ -124,6 +128,8 @@ def main():
                 s_logits, e_logits = ddp(data)
                 loss = (F.cross_entropy(s_logits, target_s)
                         + F.cross_entropy(e_logits, target_e))
+            elif is_mnist:
+                loss = F.nll_loss(ddp(data), target)
             else:
                 out = ddp(data)
                 loss = F.cross_entropy(out, target)
