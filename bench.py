@@ -49,6 +49,8 @@ def parse_args():
                    choices=["bf16", "fp32"])
     p.add_argument("--seq-len", type=int, default=384,
                    help="sequence length for bert-large")
+    p.add_argument("--fused-optimizer", action="store_true",
+                   help="use the single-kernel FusedSGD (HIP fused step)")
     p.add_argument("--no-channels-last", action="store_true",
                    help="disable NHWC layout for conv models (NHWC is the "
                         "MIOpen fast path on MI355X)")
@@ -89,6 +91,11 @@ def main():
         optimizer = QAdamOptimizer(model.parameters(), lr=1e-4,
                                    warmup_steps=max(args.warmup, 10))
         algorithm = GlobalAlgorithmRegistry.get("qadam")(optimizer)
+    elif args.fused_optimizer:
+        from bagua_amd.contrib import FusedSGD
+
+        optimizer = FusedSGD(model.parameters(), lr=0.01, momentum=0.9)
+        algorithm = GlobalAlgorithmRegistry.get(args.algorithm)()
     else:
         optimizer = torch.optim.SGD(model.parameters(), lr=0.01,
                                     momentum=0.9)
