@@ -104,8 +104,10 @@ class BaguaDistributedDataParallel:
 
         # speed metrics for autotune scoring
         # (reference: bagua_distributed.py:113-131)
+        from ..utils import StatisticalAverage
+
         self._speed_metrics_start = None
-        self._speed_samples = []
+        self._speed_avg = StatisticalAverage()
 
         self._autotune_client = None
         self._autotune_completed = env.get_autotune_level() == 0
@@ -320,8 +322,7 @@ class BaguaDistributedDataParallel:
             elapsed = time.time() - self._speed_metrics_start
             total_bytes = sum(b.bytes for b in self.bagua_buckets)
             if elapsed > 0:
-                self._speed_samples.append(total_bytes / elapsed)
-                del self._speed_samples[:-64]
+                self._speed_avg.record(total_bytes / elapsed)
             self._speed_metrics_start = None
 
     def _register_optimizer_hooks(self):
@@ -381,8 +382,7 @@ class BaguaDistributedDataParallel:
         if self.bagua_train_step_counter % CYCLE != 1 \
                 or self.bagua_train_step_counter <= 1:
             return
-        speed = (sum(self._speed_samples) / len(self._speed_samples)
-                 if self._speed_samples else 0.0)
+        speed = self._speed_avg.get(60.0)
         self._autotune_client.report_metrics(
             model_name=self.bagua_module_name,
             rank=env.get_rank(),

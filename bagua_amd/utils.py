@@ -1,0 +1,81 @@
+"""Misc utilities (reference: bagua/torch_api/utils.py:127-244 and the
+Rust show_version, lib.rs:103-123)."""
+
+import sys
+import time
+from typing import List, Tuple
+
+import torch
+
+
+def show_version():
+    """Print build / runtime versions (reference: lib.rs:103-123)."""
+    import bagua_amd
+
+    lines = [
+        "bagua_amd %s" % bagua_amd.__version__,
+        "python    %s" % sys.version.split()[0],
+        "torch     %s" % torch.__version__,
+        "ROCm/HIP  %s" % getattr(torch.version, "hip", None),
+        "GPU       %s" % (torch.cuda.get_device_name(0)
+                          if torch.cuda.is_available() else "none"),
+    ]
+    try:
+        from bagua_amd.ops import native
+
+        lines.append("native    %s" % ("loaded" if native.available()
+                                       else "unavailable"))
+    except Exception:  # noqa: BLE001
+        lines.append("native    error")
+    print("\n".join(lines))
+    return lines
+
+
+def flatten(tensors: List[torch.Tensor]) -> torch.Tensor:
+    return torch.cat([t.reshape(-1) for t in tensors])
+
+
+def unflatten(flat: torch.Tensor, tensors: List[torch.Tensor]
+              ) -> List[torch.Tensor]:
+    out, offset = [], 0
+    for t in tensors:
+        out.append(flat.narrow(0, offset, t.numel()).view_as(t))
+        offset += t.numel()
+    return out
+
+
+def to_bagua_datatype(dtype: torch.dtype) -> str:
+    return {
+        torch.float32: "f32",
+        torch.float16: "f16",
+        torch.bfloat16: "bf16",
+        torch.uint8: "u8",
+        torch.int64: "i64",
+    }[dtype]
+
+
+class StatisticalAverage:
+    """Exponential-window throughput average
+    (reference: utils.py:127-244): ``record(value)`` time-stamped samples;
+    ``get(window_s)`` returns the mean of samples within the window."""
+
+    def __init__(self, max_records: int = 128):
+        self.records: List[Tuple[float, float]] = []
+        self.max_records = max_records
+
+    def record(self, value: float, now: float = None):
+        now = time.time() if now is None else now
+        self.records.append((now, float(value)))
+        del self.records[:-self.max_records]
+
+    def get(self, window_s: float = 60.0, now: float = None) -> float:
+        now = time.time() if now is None else now
+        vals = [v for t, v in self.records if now - t <= window_s]
+        if not vals:
+            return 0.0
+        return sum(vals) / len(vals)
+
+    def total_recording_time(self) -> float:
+        if len(self.records) < 2:
+            return 0.0
+        return self.records[-1][0] - self.records[0][0]
