@@ -178,3 +178,72 @@ def test_decentralized_golden_model():
     for rank in range(nprocs):
         assert torch.allclose(out[rank], golden[rank], atol=1e-5), (
             "rank %d deviates from golden decentralized trajectory" % rank)
+
+
+def _worker_shift_one_golden(rank, nprocs, steps):
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.decentralized import (
+        DecentralizedAlgorithm,
+    )
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)
+    model = Net()
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=DecentralizedAlgorithm(peer_selection_mode="shift_one"))
+    for step in range(steps):
+        data, target = _make_data(rank, step)
+        optimizer.zero_grad()
+        loss = F.mse_loss(ddp(data), target)
+        loss.backward()
+        optimizer.step()
+    flat = _flat_params(model)
+    bagua_amd.deinit_process_group()
+    return flat
+
+
+def _simulate_shift_one(nprocs, steps, lr=0.05):
+    """Golden: per step, each rank pairs via the half-ring formula and
+    averages weights with its peer, then applies local grads."""
+    torch.manual_seed(13)
+    models = [Net() for _ in range(nprocs)]
+    with torch.no_grad():
+        for m in models[1:]:
+            for p, p0 in zip(m.parameters(), models[0].parameters()):
+                p.copy_(p0)
+    for step in range(steps):
+        grads = []
+        for rank, m in enumerate(models):
+            data, target = _make_data(rank, step)
+            m.zero_grad()
+            F.mse_loss(m(data), target).backward()
+            grads.append([p.grad.clone() for p in m.parameters()])
+        # pairing (reference formula,
+        # decentralized_full_precision_synchronous.rs:79-85)
+        n = nprocs
+        peers = {}
+        for rank in range(n):
+            if rank < n // 2:
+                peers[rank] = ((step + rank) % ((n + 1) // 2)) + n // 2
+            else:
+                peers[rank] = (rank - n // 2 - step) % (n // 2)
+        with torch.no_grad():
+            olds = [[p.detach().clone() for p in m.parameters()]
+                    for m in models]
+            for rank, m in enumerate(models):
+                peer = peers[rank]
+                for p, mine, theirs, g in zip(m.parameters(), olds[rank],
+                                              olds[peer], grads[rank]):
+                    p.copy_((mine + theirs) / 2 - lr * g)
+    return [_flat_params(m) for m in models]
+
+
+def test_shift_one_golden_4ranks():
+    nprocs, steps = 4, 3
+    out = run_multi_process(nprocs, _worker_shift_one_golden, args=(steps,))
+    golden = _simulate_shift_one(nprocs, steps)
+    for rank in range(nprocs):
+        assert torch.allclose(out[rank], golden[rank], atol=1e-5), (
+            "rank %d deviates from golden shift_one trajectory" % rank)

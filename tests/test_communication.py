@@ -135,3 +135,29 @@ def test_subgroup(nprocs):
     results = run_multi_process(nprocs, _worker_subgroup)
     for res in results:
         assert torch.allclose(res, torch.ones(4) * 3)
+
+
+def _worker_alltoall_v(rank, nprocs):
+    import bagua_amd
+
+    bagua_amd.init_process_group()
+    # rank r sends (p+1) elements of value r to each peer p
+    send_counts = [p + 1 for p in range(nprocs)]
+    send_displs = [sum(send_counts[:p]) for p in range(nprocs)]
+    send = torch.cat([torch.ones(c) * rank for c in send_counts])
+    recv_counts = [rank + 1] * nprocs
+    recv_displs = [sum(recv_counts[:p]) for p in range(nprocs)]
+    recv = torch.zeros(sum(recv_counts))
+    bagua_amd.alltoall_v(send, send_counts, send_displs,
+                         recv, recv_counts, recv_displs)
+    bagua_amd.deinit_process_group()
+    return recv
+
+
+def test_alltoall_v():
+    nprocs = 2
+    out = run_multi_process(nprocs, _worker_alltoall_v)
+    for rank, recv in enumerate(out):
+        expect = torch.cat([torch.ones(rank + 1) * p
+                            for p in range(nprocs)])
+        assert torch.allclose(recv, expect)
