@@ -37,10 +37,10 @@ void bagua_reduce_chunk_launch(int dtype, void* x, int num_chunks,
                                int target_chunk, int average, size_t chunk,
                                hipStream_t stream);
 void bagua_compress_launch(int dtype, const void* x, uint8_t* out,
-                           uint32_t* scratch, size_t chunk,
-                           size_t chunk_stride, int num_chunks_total,
-                           int chunk_begin, int chunk_count,
-                           hipStream_t stream);
+                           uint32_t* scratch, uint32_t* partials,
+                           size_t chunk, size_t chunk_stride,
+                           int num_chunks_total, int chunk_begin,
+                           int chunk_count, hipStream_t stream);
 void bagua_decompress_launch(int dtype, const uint8_t* in, void* x,
                              size_t chunk, size_t chunk_stride,
                              int chunk_begin, int chunk_count,
@@ -372,7 +372,8 @@ class BucketExecutor {
     bool hierarchical = false;
     at::Tensor wire;       // compressed wire buffer
     at::Tensor wire_tmp;   // alltoall bounce
-    at::Tensor scratch;    // minmax scratch (uint32 x 2*chunks)
+    at::Tensor scratch;    // minmax result (uint32 x 2*chunks)
+    at::Tensor partials;   // per-block minmax partials
   };
 
   BucketExecutor(std::shared_ptr<Communicator> global,
@@ -414,6 +415,8 @@ class BucketExecutor {
         b.wire = at::empty({stride * n}, flat.options().dtype(at::kByte));
         b.wire_tmp = at::empty_like(b.wire);
         b.scratch = at::empty({2 * n}, flat.options().dtype(at::kInt));
+        b.partials = at::empty({2 * 512 * n},
+                               flat.options().dtype(at::kInt));
       }
     }
     buckets_.push_back(std::move(b));
@@ -501,9 +504,10 @@ class BucketExecutor {
     int dt = kernel_dtype(b.flat);
     auto* wire = (uint8_t*)b.wire.data_ptr();
     auto* scratch = (uint32_t*)b.scratch.data_ptr();
+    auto* partials = (uint32_t*)b.partials.data_ptr();
 
-    bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, chunk,
-                          stride, n, 0, n, stream_);
+    bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, partials,
+                          chunk, stride, n, 0, n, stream_);
     comm->alltoall(b.wire, b.wire_tmp);
     std::swap(b.wire, b.wire_tmp);
     wire = (uint8_t*)b.wire.data_ptr();
@@ -511,8 +515,8 @@ class BucketExecutor {
                             n, stream_);
     bagua_reduce_chunk_launch(dt, b.flat.data_ptr(), n, rank,
                               b.average ? 1 : 0, chunk, stream_);
-    bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, chunk,
-                          stride, n, rank, 1, stream_);
+    bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, partials,
+                          chunk, stride, n, rank, 1, stream_);
     // in-place allgather of the rank's wire chunk
     {
       char* base = (char*)wire;
@@ -634,9 +638,12 @@ static void compress_chunked(at::Tensor flat, at::Tensor out,
                            flat.options().dtype(at::kInt));
   int begin = target_chunk >= 0 ? (int)target_chunk : 0;
   int count = target_chunk >= 0 ? 1 : (int)num_chunks;
+  auto partials = at::empty({2 * 512 * count},
+                            flat.options().dtype(at::kInt));
   bagua_compress_launch(kernel_dtype(flat), flat.data_ptr(),
                         (uint8_t*)out.data_ptr(),
-                        (uint32_t*)scratch.data_ptr(), chunk, stride,
+                        (uint32_t*)scratch.data_ptr(),
+                        (uint32_t*)partials.data_ptr(), chunk, stride,
                         (int)num_chunks, begin, count, current_stream());
 }
 

@@ -179,11 +179,42 @@ __device__ __forceinline__ float ordered_to_f32(uint32_t u) {
   return __uint_as_float(u);
 }
 
-__global__ void minmax_init_kernel(uint32_t* scratch, int num_chunks) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c < num_chunks) {
-    scratch[2 * c] = 0xFFFFFFFFu;  // min slot (ordered +inf)
-    scratch[2 * c + 1] = 0u;       // max slot (ordered -inf)
+#define MINMAX_MAX_BLOCKS 512
+
+// stage 2: one block per chunk reduces the per-block partials (no
+// atomics — 512 blocks contending on one address serialized ~100us)
+__global__ void minmax_finalize_kernel(const uint32_t* __restrict__
+                                       partials, int nblocks,
+                                       int chunk_begin,
+                                       uint32_t* __restrict__ scratch) {
+  const int c = chunk_begin + blockIdx.x;
+  const uint32_t* p = partials + (size_t)blockIdx.x * 2 * MINMAX_MAX_BLOCKS;
+  uint32_t mn = 0xFFFFFFFFu, mx = 0u;
+  for (int i = threadIdx.x; i < nblocks; i += blockDim.x) {
+    mn = min(mn, p[2 * i]);
+    mx = max(mx, p[2 * i + 1]);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    mn = min(mn, (uint32_t)__shfl_down((int)mn, off, 64));
+    mx = max(mx, (uint32_t)__shfl_down((int)mx, off, 64));
+  }
+  __shared__ uint32_t smn[BLOCK / 64], smx[BLOCK / 64];
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  if (lane == 0) {
+    smn[wave] = mn;
+    smx[wave] = mx;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+#pragma unroll
+    for (int w = 1; w < BLOCK / 64; ++w) {
+      mn = min(mn, smn[w]);
+      mx = max(mx, smx[w]);
+    }
+    scratch[2 * c] = mn;
+    scratch[2 * c + 1] = mx;
   }
 }
 
@@ -191,13 +222,17 @@ __global__ void minmax_init_kernel(uint32_t* scratch, int num_chunks) {
 // wave shuffle + LDS across 4 waves, one atomic pair per block.
 template <typename T>
 __global__ void minmax_kernel(const T* __restrict__ x, size_t chunk,
-                              int chunk_begin, uint32_t* scratch) {
+                              int chunk_begin, uint32_t* partials) {
   constexpr int V = Vec16<T>::N;
   using VT = Vec16<T>;
   const int c = chunk_begin + blockIdx.y;
   const T* __restrict__ src = x + (size_t)c * chunk;
 
-  float lmin = INFINITY, lmax = -INFINITY;
+  // 4 independent accumulator pairs: a single (lmin,lmax) pair makes a
+  // serial FP dependency chain per thread and the kernel goes
+  // latency-bound (~1.2 TB/s measured); 4 chains restore ILP.
+  float lmn[4] = {INFINITY, INFINITY, INFINITY, INFINITY};
+  float lmx[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
   const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   const size_t nv = chunk / V;
@@ -207,15 +242,17 @@ __global__ void minmax_kernel(const T* __restrict__ x, size_t chunk,
 #pragma unroll
     for (int k = 0; k < V; ++k) {
       float f = to_f(a.v[k]);
-      lmin = fminf(lmin, f);
-      lmax = fmaxf(lmax, f);
+      lmn[k & 3] = fminf(lmn[k & 3], f);
+      lmx[k & 3] = fmaxf(lmx[k & 3], f);
     }
   }
   for (size_t i = nv * V + tid; i < chunk; i += stride) {
     float f = to_f(src[i]);
-    lmin = fminf(lmin, f);
-    lmax = fmaxf(lmax, f);
+    lmn[0] = fminf(lmn[0], f);
+    lmx[0] = fmaxf(lmx[0], f);
   }
+  float lmin = fminf(fminf(lmn[0], lmn[1]), fminf(lmn[2], lmn[3]));
+  float lmax = fmaxf(fmaxf(lmx[0], lmx[1]), fmaxf(lmx[2], lmx[3]));
 
   // wave64 shuffle reduce
 #pragma unroll
@@ -237,8 +274,10 @@ __global__ void minmax_kernel(const T* __restrict__ x, size_t chunk,
       lmin = fminf(lmin, smin[w]);
       lmax = fmaxf(lmax, smax[w]);
     }
-    atomicMin(&scratch[2 * c], f32_to_ordered(lmin));
-    atomicMax(&scratch[2 * c + 1], f32_to_ordered(lmax));
+    // per-block partial; chunk-row-local block index
+    uint32_t* p = partials + (size_t)blockIdx.y * 2 * MINMAX_MAX_BLOCKS;
+    p[2 * blockIdx.x] = f32_to_ordered(lmin);
+    p[2 * blockIdx.x + 1] = f32_to_ordered(lmax);
   }
 }
 
@@ -446,21 +485,21 @@ void bagua_reduce_chunk_launch(int dtype, void* x, int num_chunks,
 
 // scratch: uint32[2*num_chunks_total]; out: chunked wire buffer
 void bagua_compress_launch(int dtype, const void* x, uint8_t* out,
-                           uint32_t* scratch, size_t chunk,
-                           size_t chunk_stride, int num_chunks_total,
-                           int chunk_begin, int chunk_count,
-                           hipStream_t stream) {
-  hipLaunchKernelGGL(minmax_init_kernel,
-                     dim3((num_chunks_total + BLOCK - 1) / BLOCK),
-                     dim3(BLOCK), 0, stream, scratch, num_chunks_total);
+                           uint32_t* scratch, uint32_t* partials,
+                           size_t chunk, size_t chunk_stride,
+                           int num_chunks_total, int chunk_begin,
+                           int chunk_count, hipStream_t stream) {
   int gx = grid_for(chunk / 8 + 1);
-  if (gx > 512) gx = 512;
+  if (gx > MINMAX_MAX_BLOCKS) gx = MINMAX_MAX_BLOCKS;
   dim3 grid(gx, chunk_count);
   switch (dtype) {
     case 0:
       hipLaunchKernelGGL((minmax_kernel<float>), grid, dim3(BLOCK), 0,
                          stream, (const float*)x, chunk, chunk_begin,
-                         scratch);
+                         partials);
+      hipLaunchKernelGGL(minmax_finalize_kernel, dim3(chunk_count),
+                         dim3(BLOCK), 0, stream, partials, gx,
+                         chunk_begin, scratch);
       hipLaunchKernelGGL((quantize_kernel<float>), grid, dim3(BLOCK), 0,
                          stream, (const float*)x, chunk, chunk_begin,
                          chunk_stride, scratch, out);
@@ -468,7 +507,10 @@ void bagua_compress_launch(int dtype, const void* x, uint8_t* out,
     case 1:
       hipLaunchKernelGGL((minmax_kernel<__half>), grid, dim3(BLOCK), 0,
                          stream, (const __half*)x, chunk, chunk_begin,
-                         scratch);
+                         partials);
+      hipLaunchKernelGGL(minmax_finalize_kernel, dim3(chunk_count),
+                         dim3(BLOCK), 0, stream, partials, gx,
+                         chunk_begin, scratch);
       hipLaunchKernelGGL((quantize_kernel<__half>), grid, dim3(BLOCK), 0,
                          stream, (const __half*)x, chunk, chunk_begin,
                          chunk_stride, scratch, out);
@@ -476,6 +518,9 @@ void bagua_compress_launch(int dtype, const void* x, uint8_t* out,
     case 2:
       hipLaunchKernelGGL((minmax_kernel<__hip_bfloat16>), grid, dim3(BLOCK),
                          0, stream, (const __hip_bfloat16*)x, chunk,
+                         chunk_begin, partials);
+      hipLaunchKernelGGL(minmax_finalize_kernel, dim3(chunk_count),
+                         dim3(BLOCK), 0, stream, partials, gx,
                          chunk_begin, scratch);
       hipLaunchKernelGGL((quantize_kernel<__hip_bfloat16>), grid,
                          dim3(BLOCK), 0, stream, (const __hip_bfloat16*)x,
