@@ -11,6 +11,12 @@ Launch:
 
 import torch
 
+import os
+import sys
+
+sys.path.insert(0, os.path.abspath(os.path.join(
+    os.path.dirname(os.path.abspath(__file__)), "..", "..")))
+
 import bagua_amd
 from bagua_amd import ReduceOp, env
 

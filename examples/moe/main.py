@@ -18,6 +18,12 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+import os
+import sys
+
+sys.path.insert(0, os.path.abspath(os.path.join(
+    os.path.dirname(os.path.abspath(__file__)), "..", "..")))
+
 import bagua_amd
 from bagua_amd import env
 from bagua_amd.checkpoint import load_checkpoint, save_checkpoint
