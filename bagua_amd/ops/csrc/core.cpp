@@ -160,14 +160,27 @@ class Communicator {
   void group_start() { NCCL_CHECK(ncclGroupStart()); }
   void group_end() { NCCL_CHECK(ncclGroupEnd()); }
 
+  // world-size-1 identity fast paths: a single-rank RCCL reduce still
+  // launches real kernels (~150us per 32 MiB bucket measured); these
+  // collectives are mathematically identity at nranks==1.
+  bool single_() const { return nranks_ == 1; }
+  void copy_(at::Tensor& dst, const at::Tensor& src) {
+    if (dst.data_ptr() != src.data_ptr())
+      HIP_CHECK(hipMemcpyAsync(dst.data_ptr(), src.data_ptr(),
+                               src.numel() * src.element_size(),
+                               hipMemcpyDeviceToDevice, stream_));
+  }
+
   // -- collectives (all enqueue on stream_, non-blocking host) ----------
   void allreduce_inplace(at::Tensor t, int op) {
+    if (single_()) return;
     check_device_contig(t);
     NCCL_CHECK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
                              nccl_dtype(t), nccl_op(op), comm_, stream_));
   }
 
   void allreduce(at::Tensor send, at::Tensor recv, int op) {
+    if (single_()) { copy_(recv, send); return; }
     check_device_contig(send);
     check_device_contig(recv);
     NCCL_CHECK(ncclAllReduce(send.data_ptr(), recv.data_ptr(), send.numel(),
@@ -175,18 +188,21 @@ class Communicator {
   }
 
   void reduce_inplace(at::Tensor t, int dst, int op) {
+    if (single_()) return;
     check_device_contig(t);
     NCCL_CHECK(ncclReduce(t.data_ptr(), t.data_ptr(), t.numel(),
                           nccl_dtype(t), nccl_op(op), dst, comm_, stream_));
   }
 
   void broadcast(at::Tensor t, int src) {
+    if (single_()) return;
     check_device_contig(t);
     NCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
                              nccl_dtype(t), src, comm_, stream_));
   }
 
   void allgather(at::Tensor send, at::Tensor recv) {
+    if (single_()) { copy_(recv, send); return; }
     check_device_contig(send);
     check_device_contig(recv);
     TORCH_CHECK(recv.numel() == send.numel() * nranks_,
@@ -196,6 +212,7 @@ class Communicator {
   }
 
   void allgather_inplace(at::Tensor t) {
+    if (single_()) return;
     check_device_contig(t);
     TORCH_CHECK(t.numel() % nranks_ == 0, "allgather_inplace size");
     size_t chunk = t.numel() / nranks_;
@@ -206,6 +223,7 @@ class Communicator {
   }
 
   void reduce_scatter(at::Tensor send, at::Tensor recv, int op) {
+    if (single_()) { copy_(recv, send); return; }
     check_device_contig(send);
     check_device_contig(recv);
     NCCL_CHECK(ncclReduceScatter(send.data_ptr(), recv.data_ptr(),
@@ -214,6 +232,7 @@ class Communicator {
   }
 
   void reduce_scatter_inplace(at::Tensor t, int op) {
+    if (single_()) return;
     check_device_contig(t);
     TORCH_CHECK(t.numel() % nranks_ == 0, "reduce_scatter_inplace size");
     size_t chunk = t.numel() / nranks_;
@@ -224,6 +243,7 @@ class Communicator {
   }
 
   void alltoall(at::Tensor send, at::Tensor recv) {
+    if (single_()) { copy_(recv, send); return; }
     check_device_contig(send);
     check_device_contig(recv);
     TORCH_CHECK(send.numel() % nranks_ == 0, "alltoall size");
@@ -233,6 +253,7 @@ class Communicator {
   }
 
   void alltoall_inplace(at::Tensor t) {
+    if (single_()) return;
     // RCCL alltoall is not in-place capable; bounce through a scratch
     // buffer from the torch caching allocator (cheap, stream-ordered).
     check_device_contig(t);
