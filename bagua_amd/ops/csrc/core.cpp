@@ -717,6 +717,88 @@ static py::bytes nccl_unique_id() {
   return py::bytes((const char*)&id, sizeof(id));
 }
 
+
+// ---------------------------------------------------------------------------
+// C ABI for non-Python hosts (reference: bagua-core-c/src/lib.rs:23-347).
+// Raw-pointer variants of the communicator surface; dtype codes:
+// 0=f32 1=f16 2=bf16 3=u8 4=i64. op codes = bagua_amd.ReduceOp values.
+// ---------------------------------------------------------------------------
+
+namespace {
+ncclDataType_t c_dtype(int d) {
+  switch (d) {
+    case 0: return ncclFloat32;
+    case 1: return ncclFloat16;
+    case 2: return ncclBfloat16;
+    case 3: return ncclUint8;
+    case 4: return ncclInt64;
+    default: return ncclFloat32;
+  }
+}
+}  // namespace
+
+extern "C" {
+
+void* bagua_comm_create(int rank, int nranks, int device,
+                        uintptr_t stream, const char* uid_bytes,
+                        size_t uid_len) {
+  try {
+    return new Communicator(rank, nranks, device, stream,
+                            std::string(uid_bytes, uid_len));
+  } catch (...) {
+    return nullptr;
+  }
+}
+
+void bagua_comm_destroy(void* comm) {
+  delete (Communicator*)comm;
+}
+
+int bagua_comm_rank(void* comm) { return ((Communicator*)comm)->rank(); }
+int bagua_comm_nranks(void* comm) {
+  return ((Communicator*)comm)->nranks();
+}
+void bagua_comm_abort(void* comm) { ((Communicator*)comm)->abort(); }
+
+int bagua_comm_allreduce_inplace(void* comm, void* ptr, size_t numel,
+                                 int dtype, int op, uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  if (c->nranks() == 1) return 0;
+  return ncclAllReduce(ptr, ptr, numel, c_dtype(dtype), nccl_op(op),
+                       c->raw(), (hipStream_t)stream) == ncclSuccess
+             ? 0 : -1;
+}
+
+int bagua_comm_broadcast(void* comm, void* ptr, size_t numel, int dtype,
+                         int src, uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  if (c->nranks() == 1) return 0;
+  return ncclBroadcast(ptr, ptr, numel, c_dtype(dtype), src, c->raw(),
+                       (hipStream_t)stream) == ncclSuccess ? 0 : -1;
+}
+
+int bagua_comm_allgather_inplace(void* comm, void* ptr, size_t numel,
+                                 int dtype, uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  if (c->nranks() == 1) return 0;
+  size_t chunk = numel / c->nranks();
+  size_t es = dtype == 3 ? 1 : (dtype == 1 || dtype == 2) ? 2
+              : dtype == 4 ? 8 : 4;
+  char* own = (char*)ptr + c->rank() * chunk * es;
+  return ncclAllGather(own, ptr, chunk, c_dtype(dtype), c->raw(),
+                       (hipStream_t)stream) == ncclSuccess ? 0 : -1;
+}
+
+int bagua_nccl_unique_id(char* out, size_t cap) {
+  if (cap < sizeof(ncclUniqueId)) return -1;
+  ncclUniqueId id;
+  if (ncclGetUniqueId(&id) != ncclSuccess) return -1;
+  std::memcpy(out, &id, sizeof(id));
+  return (int)sizeof(id);
+}
+
+}  // extern "C"
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "bagua_amd native core (RCCL over xGMI + CDNA4 kernels)";
 

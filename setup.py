@@ -34,4 +34,10 @@ setup(
     packages=find_packages(include=["bagua_amd", "bagua_amd.*"]),
     ext_modules=[ext],
     cmdclass={"build_ext": cpp_extension.BuildExtension},
+    entry_points={
+        "console_scripts": [
+            "baguarun = bagua_amd.distributed.baguarun:main",
+            "bagua_sys_perf = bagua_amd.distributed.sys_perf:main",
+        ],
+    },
 )

@@ -125,3 +125,52 @@ def test_train_steps_other_algorithms(algo, kwargs):
         ddp.inner.bagua_algorithm.abort(ddp)
     torch.cuda.synchronize()
     assert all(torch.isfinite(torch.tensor(losses)))
+
+
+@requires_gpu
+@pytest.mark.parametrize("algo", ["gradient_allreduce", "bytegrad"])
+def test_native_executor_matches_python_executor(algo):
+    """The C++ BucketExecutor and the torch executor must produce the
+    same training trajectory (world 1: quantization math included)."""
+    _setup_env()
+    import os
+
+    import bagua_amd
+    from bagua_amd.models import MnistNet
+    from bagua_amd.parallel.algorithms import GlobalAlgorithmRegistry
+
+    torch.cuda.set_device(0)
+    bagua_amd.init_process_group()
+
+    def run(native: bool):
+        os.environ["BAGUA_NATIVE_SCHEDULER"] = "1" if native else "0"
+        torch.manual_seed(7)
+        model = MnistNet().cuda()
+        optimizer = torch.optim.SGD(model.parameters(), lr=0.01)
+        ddp = bagua_amd.DistributedDataParallel(
+            model, optimizers=[optimizer],
+            algorithm=GlobalAlgorithmRegistry.get(algo)())
+        if native:
+            assert any(getattr(b, "_native_idx", None) is not None
+                       for b in ddp.inner.bagua_buckets)
+        else:
+            assert all(getattr(b, "_native_idx", None) is None
+                       for b in ddp.inner.bagua_buckets)
+        for step in range(4):
+            torch.manual_seed(100 + step)
+            data = torch.randn(16, 1, 28, 28, device="cuda")
+            target = torch.randint(0, 10, (16,), device="cuda")
+            optimizer.zero_grad()
+            loss = F.nll_loss(ddp(data), target)
+            loss.backward()
+            optimizer.step()
+        torch.cuda.synchronize()
+        os.environ["BAGUA_NATIVE_SCHEDULER"] = "1"
+        return torch.cat([p.detach().reshape(-1).float().cpu()
+                          for p in model.parameters()])
+
+    a = run(native=True)
+    b = run(native=False)
+    assert torch.allclose(a, b, atol=1e-6), (
+        "native executor deviates from python executor (max diff %g)"
+        % (a - b).abs().max().item())

@@ -143,3 +143,30 @@ def test_functional_allreduce_autograd():
         assert torch.allclose(y, torch.ones(4) * 3)  # 1 + 2
         # backward allreduces the ones-grad -> 2, scaled by (rank+1)
         assert torch.allclose(grad, torch.ones(4) * 2 * (rank + 1))
+
+
+def _worker_duplicate_detection(rank, nprocs):
+    import bagua_amd
+    from bagua_amd.backend import BaguaBackend
+    from bagua_amd.bucket import BaguaBucket
+    from bagua_amd.tensor import ensure_bagua_tensor
+
+    bagua_amd.init_process_group()
+    group = bagua_amd.communication._get_default_group()
+    backend = BaguaBackend(group)
+    t = torch.randn(8)
+    b1 = BaguaBucket([ensure_bagua_tensor(t, "x")], "b1", flatten=True)
+    t2 = torch.randn(8)
+    b2 = BaguaBucket([ensure_bagua_tensor(t2, "x")], "b2", flatten=True)
+    try:
+        backend.register_ordered_buckets([b1, b2])
+        raised = False
+    except ValueError:
+        raised = True
+    bagua_amd.deinit_process_group()
+    return raised
+
+
+def test_duplicate_tensor_name_rejected():
+    out = run_multi_process(1, _worker_duplicate_detection)
+    assert out[0], "duplicate tensor registration was not rejected"
