@@ -20,7 +20,7 @@ Re-design of the reference's Rust BaguaCommBackend
 import logging
 import os
 import time
-from typing import List, Optional
+from typing import List
 
 import torch
 

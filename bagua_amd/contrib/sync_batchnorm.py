@@ -7,8 +7,6 @@ backward reduces (sum_dy, sum_dy_xmu). Works on CPU/gloo and GPU/RCCL
 through the same communication layer.
 """
 
-from typing import Optional
-
 import torch
 import torch.nn.functional as F
 from torch.nn.modules.batchnorm import _BatchNorm

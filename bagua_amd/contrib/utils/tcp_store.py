@@ -14,7 +14,7 @@ import socket
 import socketserver
 import struct
 import threading
-from typing import Dict, List, Optional, Union
+from typing import Dict, List, Union
 
 import msgpack
 

@@ -18,7 +18,7 @@ PyTorch on ROCm:
 import logging
 import os
 import time
-from typing import Callable, Dict, List, Optional
+from typing import Dict, List, Optional
 
 import torch
 from torch.autograd import Variable
@@ -32,7 +32,7 @@ from ..communication import (
     broadcast_object,
     get_backend,
 )
-from ..defines import BaguaHyperparameter, TensorDeclaration, TensorDtype, dtype_bytes
+from ..defines import BaguaHyperparameter, TensorDeclaration, TensorDtype
 from ..tensor import BaguaTensor, ensure_bagua_tensor
 from .algorithms.base import Algorithm
 

@@ -9,7 +9,7 @@ exchanges (fully-connected point-to-point topology).
 """
 
 import math
-from typing import Callable, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 import torch.distributed as dist

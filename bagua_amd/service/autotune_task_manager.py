@@ -10,7 +10,6 @@ import csv
 import logging
 import os
 import time
-from collections import defaultdict
 from typing import Dict, List, Optional, Tuple
 
 from ..defines import (

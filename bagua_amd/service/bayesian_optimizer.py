@@ -13,7 +13,7 @@ and xGMI favors large fused buckets, so bucket_size_2p ranges [20, 31]
 """
 
 import random
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 INT_PARAMS = {"bucket_size_2p": (20, 31)}
 BOOL_PARAMS = ["is_hierarchical_reduce"]
