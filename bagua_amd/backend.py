@@ -194,6 +194,22 @@ class BaguaBackend:
         """Out-of-band execution (async model average loop)."""
         self._execute(bucket)
 
+    def flush_unready(self, callback):
+        """Mark every tensor of the still-pending buckets ready via
+        ``callback`` (unused-parameter support). Buckets before
+        ``_queue_idx`` already executed this iteration; at a wrapped
+        queue (idx 0) with no ready tensor there is nothing pending."""
+        if self._queue_idx == 0 and not any(
+                t.ready for b in self.ordered_buckets for t in b.tensors):
+            return
+        # snapshot the unready set first: marking the last tensor of a
+        # bucket executes it and RESETS its ready flags, which must not
+        # re-enqueue tensors we already visited
+        plan = [t for b in self.ordered_buckets[self._queue_idx:]
+                for t in b.tensors if not t.ready]
+        for t in plan:
+            callback(t)
+
     # ------------------------------------------------------------------
     def wait_pending_comm_ops(self) -> int:
         """Make the current compute stream wait on all scheduled comm

@@ -157,11 +157,10 @@ class BaguaDistributedDataParallel:
     def _bagua_broadcast_parameters(self):
         comm = self.process_group.get_global_communicator()
         module_states = []
-        for _, p in sorted(self.module.state_dict().items()):
-            if isinstance(p, torch.Tensor) and p.numel() > 0 \
-                    and p.dtype.is_floating_point:
-                module_states.append(p.data)
-            elif isinstance(p, torch.Tensor) and p.numel() > 0:
+        for name, p in sorted(self.module.state_dict().items()):
+            if name in self.parameters_to_ignore:
+                continue
+            if isinstance(p, torch.Tensor) and p.numel() > 0:
                 module_states.append(p.data)
         if module_states:
             broadcast_coalesced(module_states, src=0, comm=comm)
@@ -317,6 +316,20 @@ class BaguaDistributedDataParallel:
             self._bagua_autograd_hook_handles.append(handle)
 
     def _real_post_backward_hook(self):
+        if self.find_unused_parameters:
+            # parameters whose grad hook never fired this iteration (not
+            # part of the autograd graph) would block their bucket — and
+            # the in-order queue behind it — forever. Mark them ready with
+            # their existing (zero/stale) grads so the schedule flushes.
+            # torch DDP traverses the graph to find unused params; the
+            # ready-flag state already tells us for free.
+            backward_hook = self.bagua_algorithm.init_backward_hook(self)
+
+            def flush(bt):
+                bt.repair_bucket_view()
+                backward_hook(bt.name, bt.proxy)
+
+            self.bagua_backend.flush_unready(flush)
         self.bagua_algorithm.init_post_backward_hook(self)()
         if self._speed_metrics_start is not None:
             elapsed = time.time() - self._speed_metrics_start

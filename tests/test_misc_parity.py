@@ -170,3 +170,83 @@ def _worker_duplicate_detection(rank, nprocs):
 def test_duplicate_tensor_name_rejected():
     out = run_multi_process(1, _worker_duplicate_detection)
     assert out[0], "duplicate tensor registration was not rejected"
+
+
+class PartiallyUsedNet(nn.Module):
+    """Second branch unused when flag is off (unused-parameter case)."""
+
+    def __init__(self):
+        super().__init__()
+        self.a = nn.Linear(6, 3)
+        self.b = nn.Linear(6, 3)  # never used
+
+    def forward(self, x):
+        return self.a(x)
+
+
+def _worker_unused_params(rank, nprocs):
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)
+    model = PartiallyUsedNet()
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm(),
+        find_unused_parameters=True)
+    for step in range(4):
+        torch.manual_seed(900 + rank * 7 + step)
+        optimizer.zero_grad()
+        loss = F.mse_loss(ddp(torch.randn(4, 6)), torch.randn(4, 3))
+        loss.backward()
+        optimizer.step()
+    flat = torch.cat([p.detach().reshape(-1)
+                      for p in model.parameters()])
+    bagua_amd.deinit_process_group()
+    return flat
+
+
+def test_find_unused_parameters():
+    """Unused parameters must not wedge the in-order bucket queue; used
+    parameters stay in consensus."""
+    out = run_multi_process(2, _worker_unused_params, timeout=120)
+    assert torch.equal(out[0], out[1]), "ranks diverged with unused params"
+
+
+def _worker_ignore_list(rank, nprocs):
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)  # different init on purpose
+    model = Net()
+    model._bagua_params_and_buffers_to_ignore = ["fc2.weight", "fc2.bias"]
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm())
+    names = {n for n, _ in ddp.inner.bagua_build_params()}
+    assert "fc2.weight" not in names
+    for step in range(3):
+        torch.manual_seed(700 + rank + step)
+        optimizer.zero_grad()
+        loss = F.mse_loss(ddp(torch.randn(4, 6)), torch.randn(4, 3))
+        loss.backward()
+        optimizer.step()
+    fc1 = model.fc1.weight.detach().reshape(-1)
+    fc2 = model.fc2.weight.detach().reshape(-1)
+    bagua_amd.deinit_process_group()
+    return fc1, fc2
+
+
+def test_params_to_ignore_stay_local():
+    out = run_multi_process(2, _worker_ignore_list)
+    assert torch.equal(out[0][0], out[1][0]), "synced param diverged"
+    assert not torch.equal(out[0][1], out[1][1]), (
+        "ignored param was synced")
