@@ -40,6 +40,9 @@ _default_pg: Optional["BaguaProcessGroup"] = None
 
 _autotune_server = None
 
+# bumped on deinit so re-created communicators exchange fresh RCCL ids
+_uid_epoch = 0
+
 
 class ReduceOp(enum.IntEnum):
     """Reduction operations, values matching the reference enum
@@ -113,7 +116,9 @@ class BaguaCommunicator:
 
         N.require()
         store = c10d._get_default_store()
-        key = "bagua_amd_uid_{}".format(self.name)
+        # epoch guards against stale unique ids when a process group is
+        # re-created with the same name after deinit_process_group()
+        key = "bagua_amd_uid_{}_{}".format(self.name, _uid_epoch)
         if self.rank_in_comm == 0:
             uid = N.lib().nccl_unique_id()
             store.set(key, base64.b64encode(uid).decode())
@@ -619,7 +624,8 @@ def init_process_group(store=None, rank: int = -1, world_size: int = -1):
 
 def deinit_process_group():
     """Tear down bagua state (tests)."""
-    global _default_pg, _autotune_server
+    global _default_pg, _autotune_server, _uid_epoch
+    _uid_epoch += 1
     _default_pg = None
     if _autotune_server is not None:
         _autotune_server.shutdown()
