@@ -55,6 +55,10 @@ class _FlatGroupOptimizer(torch.optim.Optimizer):
 
 
 class FusedSGD(_FlatGroupOptimizer):
+    """Single-kernel SGD over flat param groups. bf16 parameters get an
+    fp32 master copy automatically (pure-bf16 training: fwd/bwd and the
+    gradient allreduce run at half the bytes, the update stays exact)."""
+
     def __init__(self, params: Iterable, lr: float, momentum: float = 0.0,
                  dampening: float = 0.0, weight_decay: float = 0.0,
                  nesterov: bool = False):
@@ -63,8 +67,13 @@ class FusedSGD(_FlatGroupOptimizer):
         super().__init__(params, defaults)
         self._flat = [self._flatten_group(g) for g in self.param_groups]
         for g, rec in zip(self.param_groups, self._flat):
-            if rec is not None and g["momentum"] != 0:
-                rec["momentum_buffer"] = torch.zeros_like(rec["flat_w"])
+            if rec is None:
+                continue
+            if rec["flat_w"].dtype == torch.bfloat16:
+                rec["master"] = rec["flat_w"].float()
+            if g["momentum"] != 0:
+                rec["momentum_buffer"] = torch.zeros_like(
+                    rec.get("master", rec["flat_w"]))
                 rec["momentum_initialized"] = False
 
     @torch.no_grad()
@@ -80,6 +89,7 @@ class FusedSGD(_FlatGroupOptimizer):
             w, g = rec["flat_w"], rec["flat_g"]
             mu = group["momentum"]
             m = rec.get("momentum_buffer")
+            master = rec.get("master")
             if w.is_cuda and native.available() \
                     and w.dtype == torch.float32:
                 native.lib().fused_sgd_step(
@@ -87,10 +97,18 @@ class FusedSGD(_FlatGroupOptimizer):
                     group["dampening"], group["weight_decay"],
                     group["nesterov"],
                     rec.get("momentum_initialized", False))
+            elif w.is_cuda and native.available() \
+                    and w.dtype == torch.bfloat16:
+                native.lib().fused_sgd_mixed_step(
+                    w, g, master, m if m is not None else master,
+                    group["lr"], mu, group["dampening"],
+                    group["weight_decay"], group["nesterov"],
+                    rec.get("momentum_initialized", False))
             else:
-                grad = g
+                ref = master if master is not None else w
+                grad = g.float() if master is not None else g
                 if group["weight_decay"] != 0:
-                    grad = grad.add(w, alpha=group["weight_decay"])
+                    grad = grad.add(ref, alpha=group["weight_decay"])
                 if mu != 0:
                     if rec["momentum_initialized"]:
                         m.mul_(mu).add_(grad, alpha=1 - group["dampening"])
@@ -98,7 +116,9 @@ class FusedSGD(_FlatGroupOptimizer):
                         m.copy_(grad)
                     grad = grad.add(m, alpha=mu) if group["nesterov"] \
                         else m
-                w.add_(grad, alpha=-group["lr"])
+                ref.add_(grad, alpha=-group["lr"])
+                if master is not None:
+                    w.copy_(master.to(w.dtype))
             if mu != 0:
                 rec["momentum_initialized"] = True
         return loss

@@ -54,6 +54,11 @@ void bagua_fused_adam_launch(float* p, const float* g, float* m, float* v,
                              float lr, float beta1, float beta2, float eps,
                              float weight_decay, int adamw, float bc1,
                              float bc2, size_t n, hipStream_t stream);
+void bagua_fused_sgd_mixed_launch(void* p, const void* g, float* master,
+                                  float* m, float lr, float momentum,
+                                  float dampening, float weight_decay,
+                                  int nesterov, int momentum_initialized,
+                                  size_t n, hipStream_t stream);
 }
 
 #define HIP_CHECK(cmd)                                                    \
@@ -696,6 +701,22 @@ static void fused_sgd_step(at::Tensor p, at::Tensor g, at::Tensor m,
       current_stream());
 }
 
+static void fused_sgd_mixed_step(at::Tensor p, at::Tensor g,
+                                 at::Tensor master, at::Tensor m,
+                                 double lr, double momentum,
+                                 double dampening, double weight_decay,
+                                 bool nesterov, bool momentum_initialized) {
+  check_device_contig(p);
+  TORCH_CHECK(p.scalar_type() == at::kBFloat16, "params must be bf16");
+  TORCH_CHECK(master.scalar_type() == at::kFloat, "master must be f32");
+  bagua_fused_sgd_mixed_launch(
+      p.data_ptr(), g.data_ptr(), (float*)master.data_ptr(),
+      momentum != 0.0 ? (float*)m.data_ptr() : nullptr, (float)lr,
+      (float)momentum, (float)dampening, (float)weight_decay,
+      nesterov ? 1 : 0, momentum_initialized ? 1 : 0, p.numel(),
+      current_stream());
+}
+
 static void fused_adam_step(at::Tensor p, at::Tensor g, at::Tensor m,
                             at::Tensor v, int64_t step, double lr,
                             double beta1, double beta2, double eps,
@@ -851,5 +872,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decompress_chunked", &decompress_chunked);
   m.def("compressed_chunk_stride", &compressed_chunk_stride);
   m.def("fused_sgd_step", &fused_sgd_step);
+  m.def("fused_sgd_mixed_step", &fused_sgd_mixed_step);
   m.def("fused_adam_step", &fused_adam_step);
 }

@@ -659,6 +659,48 @@ void bagua_fused_sgd_launch(float* p, const float* g, float* m, float lr,
                      nesterov, momentum_initialized, n);
 }
 
+
+// bf16 weights + fp32 master + bf16 grads: the pure-bf16 training path
+// (fp32 master keeps SGD exact; bf16 params keep fwd/bwd and the grad
+// allreduce at half the bytes)
+__global__ void fused_sgd_mixed_kernel(__hip_bfloat16* __restrict__ p,
+                                       const __hip_bfloat16* __restrict__ g,
+                                       float* __restrict__ master,
+                                       float* __restrict__ m, float lr,
+                                       float momentum, float dampening,
+                                       float weight_decay, int nesterov,
+                                       int momentum_initialized, size_t n) {
+  const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = tid; i < n; i += stride) {
+    float w = master[i];
+    float grad = __bfloat162float(g[i]) + weight_decay * w;
+    float upd = grad;
+    if (momentum != 0.f) {
+      float buf = momentum_initialized
+                      ? momentum * m[i] + (1.f - dampening) * grad
+                      : grad;
+      m[i] = buf;
+      upd = nesterov ? grad + momentum * buf : buf;
+    }
+    w -= lr * upd;
+    master[i] = w;
+    p[i] = __float2bfloat16(w);
+  }
+}
+
+void bagua_fused_sgd_mixed_launch(void* p, const void* g, float* master,
+                                  float* m, float lr, float momentum,
+                                  float dampening, float weight_decay,
+                                  int nesterov, int momentum_initialized,
+                                  size_t n, hipStream_t stream) {
+  int grid = grid_for(n + 1);
+  hipLaunchKernelGGL(fused_sgd_mixed_kernel, dim3(grid), dim3(BLOCK), 0,
+                     stream, (__hip_bfloat16*)p, (const __hip_bfloat16*)g,
+                     master, m, lr, momentum, dampening, weight_decay,
+                     nesterov, momentum_initialized, n);
+}
+
 void bagua_fused_adam_launch(float* p, const float* g, float* m, float* v,
                              float lr, float beta1, float beta2, float eps,
                              float weight_decay, int adamw, float bc1,

@@ -51,6 +51,11 @@ def parse_args():
                    help="sequence length for bert-large")
     p.add_argument("--fused-optimizer", action="store_true",
                    help="use the single-kernel FusedSGD (HIP fused step)")
+    p.add_argument("--pure-bf16", action="store_true",
+                   help="bf16 parameters + fp32-master FusedSGD (no "
+                        "autocast casts; half-width gradient allreduce). "
+                        "Default ON for vgg16 (BN-free) on GPU.")
+    p.add_argument("--no-pure-bf16", action="store_true")
     p.add_argument("--no-channels-last", action="store_true",
                    help="disable NHWC layout for conv models (NHWC is the "
                         "MIOpen fast path on MI355X)")
@@ -75,6 +80,12 @@ def main():
     from bagua_amd.models import create_model
     from bagua_amd.parallel.algorithms import GlobalAlgorithmRegistry
 
+    if (args.model == "vgg16" and use_cuda and args.dtype == "bf16"
+            and args.algorithm in ("gradient_allreduce", "bytegrad",
+                                   "decentralized", "async")
+            and not args.no_pure_bf16):
+        args.pure_bf16 = True
+
     bagua_amd.init_process_group()
 
     torch.manual_seed(42)
@@ -84,6 +95,8 @@ def main():
                      and not args.no_channels_last)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
+    if args.pure_bf16:
+        model = model.to(torch.bfloat16)
 
     if args.algorithm == "qadam":
         from bagua_amd.parallel.algorithms.q_adam import QAdamOptimizer
@@ -91,7 +104,7 @@ def main():
         optimizer = QAdamOptimizer(model.parameters(), lr=1e-4,
                                    warmup_steps=max(args.warmup, 10))
         algorithm = GlobalAlgorithmRegistry.get("qadam")(optimizer)
-    elif args.fused_optimizer:
+    elif args.fused_optimizer or args.pure_bf16:
         from bagua_amd.contrib import FusedSGD
 
         optimizer = FusedSGD(model.parameters(), lr=0.01, momentum=0.9)
@@ -108,6 +121,7 @@ def main():
     # reference used the same fixed batch per iteration too)
     is_bert = args.model.startswith("bert")
     is_mnist = args.model == "mnist"
+    bf16_data = torch.bfloat16 if args.pure_bf16 else None
     if is_mnist:  # tiny CPU-testable path for the distributed plumbing
         data = torch.randn(args.batch_size, 1, 28, 28, device=device)
         target = torch.randint(0, 10, (args.batch_size,), device=device)
@@ -122,11 +136,13 @@ def main():
         data = torch.randn(args.batch_size, 3, 224, 224, device=device)
         if channels_last:
             data = data.to(memory_format=torch.channels_last)
+        if bf16_data is not None:
+            data = data.to(bf16_data)
         target = torch.randint(0, 1000, (args.batch_size,), device=device)
 
     use_bf16 = args.dtype == "bf16"
     amp_ctx = torch.autocast(device_type=device.type, dtype=torch.bfloat16,
-                             enabled=use_bf16)
+                             enabled=use_bf16 and not args.pure_bf16)
 
     def step():
         optimizer.zero_grad()
