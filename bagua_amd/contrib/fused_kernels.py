@@ -53,6 +53,38 @@ class _FlatGroupOptimizer(torch.optim.Optimizer):
                 view.copy_(p.grad.detach())
                 p.grad = view
 
+    # flat-group state lives outside optimizer.state; (de)serialize it so
+    # checkpointing (bagua_amd.checkpoint) round-trips momentum/master
+    _FLAT_STATE_KEYS = ("momentum_buffer", "master", "exp_avg",
+                        "exp_avg_sq")
+
+    def state_dict(self):
+        sd = super().state_dict()
+        sd["bagua_flat_state"] = [
+            {k: (v.clone() if isinstance(v, torch.Tensor) else v)
+             for k, v in rec.items()
+             if k in self._FLAT_STATE_KEYS
+             or k in ("momentum_initialized", "step")}
+            if rec is not None else None
+            for rec in self._flat
+        ]
+        return sd
+
+    def load_state_dict(self, state_dict):
+        flat_state = state_dict.pop("bagua_flat_state", None)
+        super().load_state_dict(state_dict)
+        if flat_state is None:
+            return
+        for rec, saved in zip(self._flat, flat_state):
+            if rec is None or saved is None:
+                continue
+            for k, v in saved.items():
+                if isinstance(v, torch.Tensor) and k in rec \
+                        and isinstance(rec[k], torch.Tensor):
+                    rec[k].copy_(v.to(rec[k].device))
+                else:
+                    rec[k] = v
+
 
 class FusedSGD(_FlatGroupOptimizer):
     """Single-kernel SGD over flat param groups. bf16 parameters get an

@@ -113,3 +113,17 @@ def test_bf16_master_kernel_matches_cpu_math():
     # bf16 fwd/bwd on different devices can differ by rounding; the
     # master-weight update itself is exact
     assert torch.allclose(wg, wc, atol=2e-2), (wg - wc).abs().max()
+
+
+def test_fused_sgd_state_dict_roundtrip():
+    m = _model()
+    o = FusedSGD(m.parameters(), lr=0.05, momentum=0.9)
+    _train(m, o, steps=3)
+    sd = o.state_dict()
+
+    m2 = copy.deepcopy(m)
+    o2 = FusedSGD(m2.parameters(), lr=0.05, momentum=0.9)
+    o2.load_state_dict(sd)
+    a = _train(m, o, steps=2)
+    b = _train(m2, o2, steps=2)
+    assert torch.equal(a.float(), b.float()), "state_dict roundtrip broke"
