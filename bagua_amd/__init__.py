@@ -61,3 +61,17 @@ from .data_parallel import (  # noqa: F401
     DistributedDataParallel,
 )
 from .distributed_module import BaguaModule, with_bagua  # noqa: F401
+
+from . import torch_api  # noqa: E402,F401  (reference-style import alias)
+
+
+def enable_logging(level=None):
+    """Configure logging from LOG_LEVEL env (reference: the Rust core's
+    tracing subscriber, bagua-core-py/src/lib.rs:542-547)."""
+    import logging
+    import os
+
+    lvl = level or os.environ.get("LOG_LEVEL", "WARNING")
+    logging.basicConfig(
+        level=getattr(logging, str(lvl).upper(), logging.WARNING),
+        format="%(asctime)s %(name)s %(levelname)s %(message)s")
