@@ -31,6 +31,11 @@ from .tensor import BaguaTensor
 
 logger = logging.getLogger(__name__)
 
+# roctx marker ranges around bucket comm (nvtx maps to roctx on ROCm);
+# BAGUA_ROCTX=0 disables if marker overhead matters
+_ROCTX = (os.environ.get("BAGUA_ROCTX", "1") == "1"
+          and torch.cuda.is_available())
+
 
 class EventPool:
     """Reusable hipEvent pool (reference: resource_pool/mod.rs:62-98)."""
@@ -164,6 +169,16 @@ class BaguaBackend:
             self._queue_idx = 0
 
     def _execute(self, bucket: BaguaBucket):
+        if _ROCTX:
+            # shows up as a marker range in rocprofv3 --marker-trace
+            torch.cuda.nvtx.range_push("bagua_bucket:%s" % bucket.name)
+        try:
+            self._execute_inner(bucket)
+        finally:
+            if _ROCTX:
+                torch.cuda.nvtx.range_pop()
+
+    def _execute_inner(self, bucket: BaguaBucket):
         if getattr(bucket, "_native_idx", None) is not None:
             events = []
             for t in bucket.tensors:
