@@ -109,8 +109,10 @@ class BaguaBucket:
         self.padding = 0
         self.backend_bucket = None  # native registration handle
         self.flatten = flatten
+        self.alignment = alignment
         self.ops: List[object] = []
         self._flat: Optional[torch.Tensor] = None
+        self._gather_buf: Optional[torch.Tensor] = None
 
         for t in self.tensors:
             t.bucket = self
@@ -148,11 +150,47 @@ class BaguaBucket:
 
     def comm_tensor(self) -> torch.Tensor:
         """The in-place communication view: the fused buffer when
-        flattened. Raises otherwise — non-flattened buckets communicate
-        per-tensor via the executor's gather path."""
-        assert self._flat is not None, (
-            "bucket %s is not flattened; comm_tensor unavailable" % self.name)
-        return self._flat
+        flattened; for non-flattened buckets a persistent gather buffer
+        (use :meth:`comm_view` so data is staged/copied back — reference:
+        datatypes/mod.rs:1011-1088 force_copy path)."""
+        if self._flat is not None:
+            return self._flat
+        if self._gather_buf is None:
+            total = sum(t.numel() for t in self.tensors)
+            if self.alignment > 1:
+                total = ((total + self.alignment - 1)
+                         // self.alignment) * self.alignment
+            e0 = self.tensors[0].materialized()
+            self._gather_buf = torch.zeros(total, dtype=e0.dtype,
+                                           device=e0.device)
+        return self._gather_buf
+
+    def comm_view(self):
+        """Context manager yielding the communication buffer; stages
+        tensors in and copies results back when the bucket is not
+        flattened (zero-copy when it is)."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def ctx():
+            buf = self.comm_tensor()
+            if self._flat is None:
+                offset = 0
+                for t in self.tensors:
+                    eff = t.materialized()
+                    buf.narrow(0, offset, eff.numel()).copy_(
+                        eff.detach().reshape(-1))
+                    offset += eff.numel()
+            yield buf
+            if self._flat is None:
+                offset = 0
+                for t in self.tensors:
+                    eff = t.tensor()
+                    eff.detach().reshape(-1).copy_(
+                        buf.narrow(0, offset, eff.numel()))
+                    offset += eff.numel()
+
+        return ctx()
 
     def check_flatten(self) -> bool:
         if self._flat is None:

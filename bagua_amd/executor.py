@@ -97,22 +97,22 @@ def execute_ops(bucket: BaguaBucket, group: BaguaProcessGroup, backend=None):
 
 def _exec_centralized(op: CentralizedSyncOp, bucket: BaguaBucket,
                       group: BaguaProcessGroup):
-    flat = bucket.comm_tensor()
-    ctx = _HierCtx(group, op.hierarchical)
-    ctx.pre(flat, average=op.average)
-    if ctx.run_inner():
-        comm = ctx.comm()
-        if op.compression is None:
-            if not op.scattergather:
-                comm.allreduce_inplace(
-                    flat, ReduceOp.AVG if op.average else ReduceOp.SUM)
+    with bucket.comm_view() as flat:
+        ctx = _HierCtx(group, op.hierarchical)
+        ctx.pre(flat, average=op.average)
+        if ctx.run_inner():
+            comm = ctx.comm()
+            if op.compression is None:
+                if not op.scattergather:
+                    comm.allreduce_inplace(
+                        flat, ReduceOp.AVG if op.average else ReduceOp.SUM)
+                else:
+                    _scattergather_sync(comm, flat, op.average)
+            elif op.compression == "MinMaxUInt8":
+                _compressed_sync(comm, flat, op.average)
             else:
-                _scattergather_sync(comm, flat, op.average)
-        elif op.compression == "MinMaxUInt8":
-            _compressed_sync(comm, flat, op.average)
-        else:
-            raise ValueError("unknown compression %r" % op.compression)
-    ctx.post(flat)
+                raise ValueError("unknown compression %r" % op.compression)
+        ctx.post(flat)
 
 
 def _scattergather_sync(comm, flat, average):
@@ -178,7 +178,12 @@ def _on_comm_stream(comm, fn):
 
 def _exec_decentralized(op: DecentralizedSyncOp, bucket: BaguaBucket,
                         group: BaguaProcessGroup):
-    flat = bucket.comm_tensor()
+    with bucket.comm_view() as flat:
+        _exec_decentralized_inner(op, flat, group)
+    op.step += 1
+
+
+def _exec_decentralized_inner(op, flat, group):
     peer = op.peer_weight.tensor()
     ctx = _HierCtx(group, op.hierarchical)
     ctx.pre(flat, average=True)
@@ -206,19 +211,18 @@ def _exec_decentralized(op: DecentralizedSyncOp, bucket: BaguaBucket,
             _on_comm_stream(comm, lambda: ops.average_inplace(peer, flat))
         else:
             raise ValueError(op.peer_selection_mode)
-    op.step += 1
 
 
 def copy_back_peer_weight(op: DecentralizedSyncOp, bucket: BaguaBucket,
                           group: BaguaProcessGroup):
     """Post-backward: install averaged weights
     (reference: decentralized_full_precision_synchronous.rs:105-124)."""
-    flat = bucket.comm_tensor()
-    ctx = _HierCtx(group, op.hierarchical)
-    if ctx.run_inner():
-        _on_comm_stream(ctx.comm(),
-                        lambda: flat.copy_(op.peer_weight.tensor()))
-    ctx.post(flat)
+    with bucket.comm_view() as flat:
+        ctx = _HierCtx(group, op.hierarchical)
+        if ctx.run_inner():
+            _on_comm_stream(ctx.comm(),
+                            lambda: flat.copy_(op.peer_weight.tensor()))
+        ctx.post(flat)
 
 
 # ---------------------------------------------------------------------------
@@ -235,7 +239,11 @@ def _exec_low_prec_decentralized(op: LowPrecisionDecentralizedSyncOp,
         L += decompress(c_left);  R += decompress(c_right)
         W  = W + decompress(c);   x = W
     """
-    flat = bucket.comm_tensor()
+    with bucket.comm_view() as flat:
+        _exec_low_prec_inner(op, flat, group)
+
+
+def _exec_low_prec_inner(op, flat, group):
     W = op.weight.tensor()
     L = op.left_peer_weight.tensor()
     R = op.right_peer_weight.tensor()
@@ -291,7 +299,11 @@ def _exec_async_model_average(op: AsyncModelAverageOp, bucket: BaguaBucket,
     ``x += reduced/n - x_copy`` is applied under the lock again so the
     training thread never sees a half-applied average.
     """
-    flat = bucket.comm_tensor()
+    with bucket.comm_view() as flat:
+        _exec_async_inner(op, flat, group)
+
+
+def _exec_async_inner(op, flat, group):
     comm = (op.group or group).get_global_communicator()
     n = comm.nranks()
 

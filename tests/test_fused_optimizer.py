@@ -70,3 +70,40 @@ def test_fuse_then_plain_step_still_works():
     loss.backward()
     o.step()  # plain step must keep working
     assert all(torch.isfinite(p).all() for p in m.parameters())
+
+
+def _worker_fused_with_ddp(rank, nprocs):
+    """Reference-documented combo: fuse_optimizer + with_bagua
+    do_flatten=False (fuse/optimizer.py docstring)."""
+    import bagua_amd
+    from bagua_amd.contrib import fuse_optimizer
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+
+    torch.manual_seed(13 + rank)
+    bagua_amd.init_process_group()
+    model = _model()
+    optimizer = fuse_optimizer(
+        torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9))
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm(),
+        gradient_as_bucket_view=False)
+    torch.manual_seed(80 + rank)
+    for _ in range(5):
+        x, y = torch.randn(6, 13), torch.randn(6, 3)
+        optimizer.zero_grad()
+        ((ddp(x) - y) ** 2).mean().backward()
+        optimizer.fuse_step()
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    bagua_amd.deinit_process_group()
+    return flat
+
+
+def test_fused_optimizer_with_ddp_no_flatten():
+    from tests.internal.multi_process import run_multi_process
+
+    out = run_multi_process(2, _worker_fused_with_ddp)
+    assert torch.equal(out[0], out[1]), "ranks diverged"
+    assert torch.isfinite(out[0]).all()

@@ -107,3 +107,32 @@ def test_no_sync_keeps_local_grads():
     out = run_multi_process(2, _worker_no_sync)
     assert not torch.equal(out[0], out[1]), (
         "grads were synced inside no_sync()")
+
+
+def _worker_no_flatten(rank, nprocs):
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)
+    model = Net()
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm(),
+        gradient_as_bucket_view=False)
+    _train(rank, ddp, optimizer)
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    bagua_amd.deinit_process_group()
+    return flat
+
+
+def test_no_flatten_matches_flatten():
+    """do_flatten=False (gather-copy comm path, used with the fused
+    optimizer) must produce the same result as the fused-view path."""
+    out = run_multi_process(2, _worker_no_flatten)
+    assert torch.equal(out[0], out[1])
+    ref = run_multi_process(2, _worker_bagua)
+    assert torch.allclose(out[0], ref[0], atol=1e-6)
