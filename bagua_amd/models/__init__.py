@@ -6,7 +6,13 @@ image, so the architectures are implemented here directly.
 """
 
 from .vgg import VGG, vgg16, vgg11, vgg13, vgg19  # noqa: F401
-from .resnet import ResNet, resnet18, resnet50  # noqa: F401
+from .resnet import (  # noqa: F401
+    ResNet,
+    resnet18,
+    resnet50,
+    resnet101,
+    resnet152,
+)
 from .mnist import MnistNet  # noqa: F401
 from .bert import BertConfig, BertForPretrainingShape, bert_large  # noqa: F401
 
@@ -20,6 +26,8 @@ def create_model(name: str, num_classes: int = 1000):
         "vgg19": vgg19,
         "resnet18": resnet18,
         "resnet50": resnet50,
+        "resnet101": resnet101,
+        "resnet152": resnet152,
     }
     if name in factory:
         return factory[name](num_classes=num_classes)
