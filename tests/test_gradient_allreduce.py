@@ -136,3 +136,65 @@ def test_no_flatten_matches_flatten():
     assert torch.equal(out[0], out[1])
     ref = run_multi_process(2, _worker_bagua)
     assert torch.allclose(out[0], ref[0], atol=1e-6)
+
+
+def _worker_grad_accumulation(rank, nprocs):
+    """no_sync accumulation then synced step must match torch DDP."""
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13)
+    model = Net()
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm())
+    for step in range(3):
+        optimizer.zero_grad()
+        with ddp.no_sync():  # micro-batch 1: local accumulation
+            torch.manual_seed(600 + rank * 17 + step * 2)
+            loss = F.mse_loss(ddp(torch.randn(4, 8)), torch.randn(4, 4))
+            loss.backward()
+        # micro-batch 2: synced
+        torch.manual_seed(601 + rank * 17 + step * 2)
+        loss = F.mse_loss(ddp(torch.randn(4, 8)), torch.randn(4, 4))
+        loss.backward()
+        optimizer.step()
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    bagua_amd.deinit_process_group()
+    return flat
+
+
+def _worker_grad_accumulation_torch(rank, nprocs):
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=nprocs)
+    torch.manual_seed(13)
+    model = Net()
+    ddp = torch.nn.parallel.DistributedDataParallel(model)
+    optimizer = torch.optim.SGD(ddp.parameters(), lr=0.05)
+    for step in range(3):
+        optimizer.zero_grad()
+        with ddp.no_sync():
+            torch.manual_seed(600 + rank * 17 + step * 2)
+            loss = F.mse_loss(ddp(torch.randn(4, 8)), torch.randn(4, 4))
+            loss.backward()
+        torch.manual_seed(601 + rank * 17 + step * 2)
+        loss = F.mse_loss(ddp(torch.randn(4, 8)), torch.randn(4, 4))
+        loss.backward()
+        optimizer.step()
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    dist.destroy_process_group()
+    return flat
+
+
+def test_gradient_accumulation_matches_torch_ddp():
+    ours = run_multi_process(2, _worker_grad_accumulation)
+    ref = run_multi_process(2, _worker_grad_accumulation_torch)
+    assert torch.equal(ours[0], ours[1])
+    assert torch.allclose(ours[0], ref[0], atol=1e-6), (
+        "grad accumulation deviates from torch DDP (max %g)"
+        % (ours[0] - ref[0]).abs().max())
