@@ -156,3 +156,39 @@ def test_dense_checkpoint_roundtrip():
             assert torch.allclose(p, ref[n])
         assert os.path.exists(
             os.path.join(tmp, "iter_0000007", "mp_rank_00_model_states.pt"))
+
+
+def test_gate_capacity_and_laux():
+    """Top-1 gate respects capacity and emits a finite aux loss."""
+    import torch as t
+
+    from bagua_amd.parallel.moe.sharded_moe import top1gating
+
+    t.manual_seed(3)
+    logits = t.randn(64, 4)
+    l_aux, combine, dispatch, counts = top1gating(
+        logits, capacity_factor=1.0, min_capacity=4)
+    assert t.isfinite(l_aux)
+    cap = combine.shape[2]
+    assert cap == max(4, 64 // 4)
+    # no expert slot double-booked
+    per_slot = dispatch.sum(dim=0)  # (experts, capacity)
+    assert int(per_slot.max()) <= 1
+    # every kept token has exactly one (expert, slot)
+    per_token = dispatch.reshape(64, -1).sum(1)
+    assert int(per_token.max()) <= 1
+
+
+def test_gate_top2_weights_normalized():
+    import torch as t
+
+    from bagua_amd.parallel.moe.sharded_moe import top2gating
+
+    t.manual_seed(4)
+    logits = t.randn(32, 4)
+    l_aux, combine, dispatch, counts = top2gating(
+        logits, capacity_factor=1.0, min_capacity=4)
+    sums = combine.reshape(32, -1).sum(1)
+    kept = sums > 0
+    assert t.all(sums[kept] <= 1.0 + 1e-5)
+    assert t.isfinite(l_aux)
