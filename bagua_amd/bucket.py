@@ -16,6 +16,7 @@ is divisible by the world size times the dtype's 32-byte alignment quantum
 construction).
 """
 
+import contextlib
 from dataclasses import dataclass, field
 from typing import Callable, List, Optional
 
@@ -169,7 +170,6 @@ class BaguaBucket:
         """Context manager yielding the communication buffer; stages
         tensors in and copies results back when the bucket is not
         flattened (zero-copy when it is)."""
-        import contextlib
 
         @contextlib.contextmanager
         def ctx():

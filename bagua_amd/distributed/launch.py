@@ -66,7 +66,6 @@ def main(args=None):
         cmd.extend(args.training_script_args)
         processes.append(subprocess.Popen(cmd, env=env))
 
-    sig_names = {2: "SIGINT", 15: "SIGTERM"}
     try:
         alive = list(processes)
         while alive:

@@ -114,7 +114,6 @@ class BaguaDistributedDataParallel:
         self._current_hp = BaguaHyperparameter(
             bucket_size=env.get_default_bucket_size())
         self._bagua_autograd_hook_handles = []
-        self._hooks_fired_this_iter = False
 
         self._install_forward_pre_hooks()
         self._bagua_init_algorithm()

@@ -15,10 +15,6 @@ and xGMI favors large fused buckets, so bucket_size_2p ranges [20, 31]
 import random
 from typing import Dict, List, Tuple
 
-INT_PARAMS = {"bucket_size_2p": (20, 31)}
-BOOL_PARAMS = ["is_hierarchical_reduce"]
-
-
 class IntParam:
     def __init__(self, val: int, space_dimension: Tuple[int, int]):
         self.val = int(val)
