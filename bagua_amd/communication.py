@@ -596,14 +596,22 @@ def init_process_group(store=None, rank: int = -1, world_size: int = -1):
         world_size = env.get_world_size()
 
     if not dist.is_initialized():
+        from datetime import timedelta
+
         backend = "nccl" if _is_cuda_job() else "gloo"
+        # collective timeout (surviving ranks of a crashed gang must fail
+        # fast so torchelastic can restart them; reference relied on its
+        # 300 s comm watchdog for the same purpose)
+        timeout = timedelta(
+            seconds=int(os.environ.get("BAGUA_PG_TIMEOUT_S", "1800")))
         if store is None:
             os.environ.setdefault("MASTER_ADDR", env.get_master_addr())
             os.environ.setdefault("MASTER_PORT", str(env.get_master_port()))
-            dist.init_process_group(backend, rank=rank, world_size=world_size)
+            dist.init_process_group(backend, rank=rank,
+                                    world_size=world_size, timeout=timeout)
         else:
             dist.init_process_group(backend, store=store, rank=rank,
-                                    world_size=world_size)
+                                    world_size=world_size, timeout=timeout)
 
     if env.get_autotune_level() > 0 and _autotune_server is None:
         from .service import autotune_service

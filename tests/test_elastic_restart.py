@@ -20,6 +20,8 @@ import torch.nn.functional as F
 
 sys.path.insert(0, {str(repr(__import__('os').getcwd()))})
 
+os.environ["BAGUA_PG_TIMEOUT_S"] = "20"  # fail fast after gang crash
+
 import bagua_amd
 from bagua_amd.checkpoint import load_checkpoint, save_checkpoint
 from bagua_amd.models import MnistNet
@@ -61,9 +63,9 @@ print("FINISHED", flush=True)
     out = subprocess.run(
         [sys.executable, "-m", "bagua_amd.distributed.run",
          "--standalone", "--nnodes=1", "--nproc-per-node=2",
-         "--local-addr", "127.0.0.1", "--max-restarts", "2",
+         "--local-addr", "127.0.0.1", "--max-restarts", "5",
          str(script)],
-        capture_output=True, text=True, timeout=300)
+        capture_output=True, text=True, timeout=420)
     assert out.returncode == 0, out.stderr[-3000:]
     assert "INJECTED_CRASH" in out.stdout
     assert "RESUMED_FROM 1" in out.stdout, out.stdout[-2000:]
