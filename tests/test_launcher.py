@@ -50,3 +50,17 @@ def test_baguarun_command_build():
     cmd = build_remote_command(args, node_rank=1, master_addr="h1")
     assert "--node_rank 1" in cmd and "--master_addr h1" in cmd
     assert "train.py --lr 0.1" in cmd
+
+
+def test_sys_perf_cli_two_procs():
+    out = subprocess.run(
+        [sys.executable, "-m", "bagua_amd.distributed.run",
+         "--standalone", "--nnodes=1", "--nproc-per-node=2",
+         "--local-addr", "127.0.0.1", "-m",
+         "bagua_amd.distributed.sys_perf",
+         "--min-bytes", "4096", "--max-bytes", "16384",
+         "--iters", "3", "--warmup", "1"],
+        capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "busbw" in out.stdout
+    assert '"world_size": 2' in out.stdout
