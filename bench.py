@@ -56,6 +56,11 @@ def parse_args():
                         "autocast casts; half-width gradient allreduce). "
                         "Default ON for vgg16 (BN-free) on GPU.")
     p.add_argument("--no-pure-bf16", action="store_true")
+    p.add_argument("--hip-graph", action="store_true",
+                   help="capture the whole training step (fwd+bwd+comm+"
+                        "optimizer) in a hipGraph after warmup and replay "
+                        "it for the timed steps (launch-overhead-free "
+                        "inner loop)")
     p.add_argument("--no-channels-last", action="store_true",
                    help="disable NHWC layout for conv models (NHWC is the "
                         "MIOpen fast path on MI355X)")
@@ -168,6 +173,39 @@ def main():
     for _ in range(args.warmup):
         step()
     barrier_sync()
+
+    if args.hip_graph and use_cuda:
+        # capture one full step; grads must keep stable pointers across
+        # replays, so zero in place instead of dropping them
+        def graph_step():
+            optimizer.zero_grad(set_to_none=False)
+            with amp_ctx:
+                if is_bert:
+                    s_logits, e_logits = ddp(data)
+                    loss = (F.cross_entropy(s_logits, target_s)
+                            + F.cross_entropy(e_logits, target_e))
+                elif is_mnist:
+                    loss = F.nll_loss(ddp(data), target)
+                else:
+                    loss = F.cross_entropy(ddp(data), target)
+            loss.backward()
+            optimizer.step()
+            return loss
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):  # graph-warmup on a side stream
+                graph_step()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            graph_step()
+        barrier_sync()
+
+        def step():  # noqa: F811 — replay path
+            g.replay()
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
