@@ -543,8 +543,8 @@ class BucketExecutor {
                               b.average ? 1 : 0, chunk, stream_);
     bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, partials,
                           chunk, stride, n, rank, 1, stream_);
-    // in-place allgather of the rank's wire chunk
-    {
+    // in-place allgather of the rank's wire chunk (identity at n==1)
+    if (n > 1) {
       char* base = (char*)wire;
       char* own = base + (int64_t)rank * stride;
       NCCL_CHECK(ncclAllGather(own, base, stride, ncclUint8,
