@@ -193,3 +193,38 @@ def test_convert_sync_batchnorm():
     conv = SyncBatchNorm.convert_sync_batchnorm(m)
     assert isinstance(conv[1], SyncBatchNorm)
     assert isinstance(conv[3][0], SyncBatchNorm)
+
+
+def test_tcp_store_concurrent_clients():
+    import threading
+
+    from bagua_amd.contrib.utils import TcpStore
+    from bagua_amd.contrib.utils.tcp_store import TcpStoreServer
+
+    server = TcpStoreServer(port=0)
+    try:
+        errors = []
+
+        def client(cid):
+            try:
+                st = TcpStore("127.0.0.1", server.port)
+                for i in range(50):
+                    st.set("c%d_k%d" % (cid, i), b"v%d" % i)
+                got = st.mget(["c%d_k%d" % (cid, i) for i in range(50)])
+                assert got == [b"v%d" % i for i in range(50)]
+                st.shutdown()
+            except Exception as e:  # noqa: BLE001
+                errors.append(e)
+
+        threads = [threading.Thread(target=client, args=(c,))
+                   for c in range(8)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=60)
+        assert not errors, errors
+        probe = TcpStore("127.0.0.1", server.port)
+        assert probe.num_keys() == 8 * 50
+        probe.shutdown()
+    finally:
+        server.shutdown()

@@ -85,3 +85,37 @@ def test_hierarchical_equals_flat():
     out = run_multi_process(2, _worker_hier_matches_flat)
     for got, expect in out:
         assert torch.allclose(got, expect, atol=1e-6)
+
+
+def test_hierarchical_qadam_consensus():
+    out = run_multi_process(2, _worker_hier_qadam)
+    assert torch.equal(out[0], out[1]), "hierarchical qadam diverged"
+
+
+def _worker_hier_qadam(rank, nprocs):
+    os.environ["NODE_RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = "0"
+
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.q_adam import (
+        QAdamAlgorithm,
+        QAdamOptimizer,
+    )
+    from tests.test_algorithms import Net, _make_data
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)
+    model = Net()
+    optimizer = QAdamOptimizer(model.parameters(), lr=1e-3, warmup_steps=4)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=QAdamAlgorithm(optimizer, hierarchical=True))
+    for step in range(8):
+        data, target = _make_data(rank, step)
+        optimizer.zero_grad()
+        loss = F.mse_loss(ddp(data), target)
+        loss.backward()
+        optimizer.step()
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    bagua_amd.deinit_process_group()
+    return flat

@@ -250,3 +250,18 @@ def test_params_to_ignore_stay_local():
     assert torch.equal(out[0][0], out[1][0]), "synced param diverged"
     assert not torch.equal(out[0][1], out[1][1]), (
         "ignored param was synced")
+
+
+def test_c_abi_symbols_present():
+    """The C ABI for non-Python hosts must export its surface
+    (reference: bagua-core-c)."""
+    import ctypes
+
+    from bagua_amd import _C
+
+    lib = ctypes.CDLL(_C.__file__)
+    for sym in ("bagua_comm_create", "bagua_comm_destroy",
+                "bagua_comm_rank", "bagua_comm_nranks", "bagua_comm_abort",
+                "bagua_comm_allreduce_inplace", "bagua_comm_broadcast",
+                "bagua_comm_allgather_inplace", "bagua_nccl_unique_id"):
+        assert hasattr(lib, sym), "missing C ABI symbol %s" % sym
