@@ -60,7 +60,11 @@ from .data_parallel import (  # noqa: F401
     BaguaDistributedDataParallel,
     DistributedDataParallel,
 )
-from .distributed_module import BaguaModule, with_bagua  # noqa: F401
+from .distributed_module import (  # noqa: F401
+    BaguaModule,
+    patch_torch_module,
+    with_bagua,
+)
 
 from . import torch_api  # noqa: E402,F401  (reference-style import alias)
 

@@ -265,3 +265,35 @@ def test_c_abi_symbols_present():
                 "bagua_comm_allreduce_inplace", "bagua_comm_broadcast",
                 "bagua_comm_allgather_inplace", "bagua_nccl_unique_id"):
         assert hasattr(lib, sym), "missing C ABI symbol %s" % sym
+
+
+def _worker_with_bagua(rank, nprocs):
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)
+    model = Net()
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+    # legacy API: function form + opt-in method patch
+    bagua_amd.patch_torch_module()
+    model = model.with_bagua([optimizer], GradientAllReduceAlgorithm())
+    assert hasattr(model, "bagua_ddp")
+    assert model.bagua_optimizers == [optimizer]
+    assert len(model.bagua_buckets) > 0
+    for step in range(4):
+        torch.manual_seed(300 + rank + step)
+        optimizer.zero_grad()
+        loss = F.mse_loss(model(torch.randn(4, 6)), torch.randn(4, 3))
+        loss.backward()
+        optimizer.step()
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    bagua_amd.deinit_process_group()
+    return flat
+
+
+def test_with_bagua_legacy_api():
+    out = run_multi_process(2, _worker_with_bagua)
+    assert torch.equal(out[0], out[1]), "with_bagua ranks diverged"
