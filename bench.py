@@ -174,6 +174,12 @@ def main():
         step()
     barrier_sync()
 
+    if args.hip_graph and args.algorithm not in ("gradient_allreduce",
+                                                 "bytegrad"):
+        raise SystemExit(
+            "--hip-graph supports the centralized synchronous algorithms "
+            "only (decentralized/async paths host-synchronize inside the "
+            "step, which cannot be captured)")
     if args.hip_graph and use_cuda:
         # capture one full step; grads must keep stable pointers across
         # replays, so zero in place instead of dropping them
