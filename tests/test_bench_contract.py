@@ -5,6 +5,8 @@ import json
 import subprocess
 import sys
 
+from tests.internal.multi_process import find_free_port
+
 
 def test_bench_distributed_cpu():
     out = subprocess.run(
@@ -36,7 +38,7 @@ def test_bench_single_process_cpu():
         capture_output=True, text=True, timeout=240,
         env={"PATH": "/usr/bin:/bin:/usr/local/bin",
              "PYTHONPATH": ".",
-             "MASTER_PORT": "29733", "HOME": "/root"})
+             "MASTER_PORT": str(find_free_port()), "HOME": "/root"})
     assert out.returncode == 0, out.stderr[-3000:]
     rec = json.loads([ln for ln in out.stdout.splitlines()
                       if ln.startswith("{")][0])
