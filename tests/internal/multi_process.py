@@ -37,9 +37,28 @@ def _worker(fn, rank, nprocs, port, args, queue):
         queue.put((rank, "error", traceback.format_exc()))
 
 
-def run_multi_process(nprocs, fn, args=(), timeout=180):
+_PORT_RACE_MARKERS = ("Address already in use", "Connection refused",
+                      "timed out", "EADDRINUSE", "connectFullMesh")
+
+
+def run_multi_process(nprocs, fn, args=(), timeout=180, retries=2):
     """Run ``fn(rank, nprocs, *args)`` in ``nprocs`` spawned processes.
-    Returns results ordered by rank. Raises on any worker failure."""
+    Returns results ordered by rank. Raises on any worker failure.
+    Rendezvous-port races (another process grabbed the ephemeral port
+    between probe and bind) are retried with a fresh port."""
+    last = None
+    for _ in range(retries + 1):
+        try:
+            return _run_once(nprocs, fn, args, timeout)
+        except RuntimeError as e:
+            msg = str(e)
+            if not any(m in msg for m in _PORT_RACE_MARKERS):
+                raise
+            last = e
+    raise last
+
+
+def _run_once(nprocs, fn, args, timeout):
     ctx = mp.get_context("spawn")
     queue = ctx.Queue()
     port = find_free_port()
