@@ -166,8 +166,11 @@ class FusedAdam(_FlatGroupOptimizer):
         self._flat = [self._flatten_group(g) for g in self.param_groups]
         for rec in self._flat:
             if rec is not None:
-                rec["exp_avg"] = torch.zeros_like(rec["flat_w"])
-                rec["exp_avg_sq"] = torch.zeros_like(rec["flat_w"])
+                if rec["flat_w"].dtype == torch.bfloat16:
+                    rec["master"] = rec["flat_w"].float()
+                ref = rec.get("master", rec["flat_w"])
+                rec["exp_avg"] = torch.zeros_like(ref)
+                rec["exp_avg_sq"] = torch.zeros_like(ref)
                 rec["step"] = 0
 
     @torch.no_grad()
@@ -183,6 +186,7 @@ class FusedAdam(_FlatGroupOptimizer):
             rec["step"] += 1
             w, g = rec["flat_w"], rec["flat_g"]
             m, v = rec["exp_avg"], rec["exp_avg_sq"]
+            master = rec.get("master")
             beta1, beta2 = group["betas"]
             if w.is_cuda and native.available() \
                     and w.dtype == torch.float32:
@@ -190,17 +194,21 @@ class FusedAdam(_FlatGroupOptimizer):
                     w, g, m, v, rec["step"], group["lr"], beta1, beta2,
                     group["eps"], group["weight_decay"], group["adamw"])
             else:
-                grad = g
+                # bf16 params: update the fp32 master, write back bf16
+                ref = master if master is not None else w
+                grad = g.float() if master is not None else g
                 if group["adamw"]:
-                    w.mul_(1 - group["lr"] * group["weight_decay"])
+                    ref.mul_(1 - group["lr"] * group["weight_decay"])
                 elif group["weight_decay"] != 0:
-                    grad = grad.add(w, alpha=group["weight_decay"])
+                    grad = grad.add(ref, alpha=group["weight_decay"])
                 m.mul_(beta1).add_(grad, alpha=1 - beta1)
                 v.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
                 bc1 = 1 - beta1 ** rec["step"]
                 bc2 = 1 - beta2 ** rec["step"]
                 denom = (v.sqrt() / (bc2 ** 0.5)).add_(group["eps"])
-                w.addcdiv_(m, denom, value=-group["lr"] / bc1)
+                ref.addcdiv_(m, denom, value=-group["lr"] / bc1)
+                if master is not None:
+                    w.copy_(master.to(w.dtype))
         return loss
 
 

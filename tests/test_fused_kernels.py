@@ -127,3 +127,25 @@ def test_fused_sgd_state_dict_roundtrip():
     a = _train(m, o, steps=2)
     b = _train(m2, o2, steps=2)
     assert torch.equal(a.float(), b.float()), "state_dict roundtrip broke"
+
+
+def test_adam_bf16_master_matches_fp32_cpu():
+    torch.manual_seed(71)
+    m32 = _model()
+    m16 = copy.deepcopy(m32).to(torch.bfloat16)
+    o32 = torch.optim.Adam(m32.parameters(), lr=1e-2)
+    o16 = FusedAdam(m16.parameters(), lr=1e-2)
+    torch.manual_seed(72)
+    for _ in range(5):
+        x = torch.randn(6, 10)
+        y = torch.randn(6, 5)
+        o32.zero_grad()
+        ((m32(x) - y) ** 2).mean().backward()
+        o32.step()
+        o16.zero_grad()
+        ((m16(x.bfloat16()) - y.bfloat16()) ** 2).mean().backward()
+        o16.step()
+    w32 = torch.cat([p.detach().reshape(-1) for p in m32.parameters()])
+    w16 = torch.cat([p.detach().float().reshape(-1)
+                     for p in m16.parameters()])
+    assert torch.allclose(w32, w16, atol=0.06), (w32 - w16).abs().max()
