@@ -308,6 +308,11 @@ __global__ void quantize_kernel(const T* __restrict__ x, size_t chunk,
     float* hdr = reinterpret_cast<float*>(dst);
     hdr[0] = mn;
     hdr[1] = mx;
+    // zero the header tail (bytes 8..31): the wire buffer is sent to
+    // peers, so uninitialized bytes would leak GPU memory contents and
+    // make the wire non-deterministic vs the zero-padded CPU oracle.
+#pragma unroll
+    for (int k = 2; k < 8; ++k) hdr[k] = 0.0f;
   }
   const T* __restrict__ src = x + (size_t)c * chunk;
   uint8_t* __restrict__ payload = dst + 32;

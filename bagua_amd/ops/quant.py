@@ -43,9 +43,12 @@ def decompress(minmax: torch.Tensor, payload: torch.Tensor,
 
 
 # ---------------------------------------------------------------------------
-# Chunked wire format (matches the native kernels / reference buffer layout:
-# per chunk, a 32-byte header holding min,max as the source dtype, then the
-# uint8 payload — datatypes/mod.rs:700-777)
+# Chunked wire format (matches the native kernels): per chunk, a 32-byte
+# header holding min,max ALWAYS as float32 (bytes 8..31 zero), then the
+# uint8 payload. Deliberate deviation from the reference's layout, which
+# stores min/max in the SOURCE dtype (datatypes/mod.rs:700-777): a fixed
+# f32 header keeps one kernel per dtype and loses no precision for
+# f16/bf16 inputs. Wire-incompatible with the reference by design.
 # ---------------------------------------------------------------------------
 
 HEADER_BYTES = 32

@@ -154,10 +154,21 @@ class BaguaDistributedDataParallel:
     # state broadcast at init (reference: bagua_distributed.py:229-323)
     # ------------------------------------------------------------------
     def _bagua_broadcast_parameters(self):
+        # MoE expert parameters are rank-local state: broadcasting rank 0's
+        # experts would collapse expert-parallel diversity (and destroy
+        # trained experts whenever the algorithm re-inits, e.g. at the QAdam
+        # warmup boundary). The reference broadcasts bagua_build_params()
+        # which excludes them (bagua_distributed.py:172, moe/utils.py:4-7);
+        # state_dict() detaches tensors so the ``.expert`` tag is lost —
+        # filter by parameter NAME instead.
+        from .moe.utils import is_moe_param
+
         comm = self.process_group.get_global_communicator()
+        moe_names = {name for name, p in self.module.named_parameters()
+                     if is_moe_param(p)}
         module_states = []
         for name, p in sorted(self.module.state_dict().items()):
-            if name in self.parameters_to_ignore:
+            if name in self.parameters_to_ignore or name in moe_names:
                 continue
             if isinstance(p, torch.Tensor) and p.numel() > 0:
                 module_states.append(p.data)
@@ -184,10 +195,20 @@ class BaguaDistributedDataParallel:
                     "cannot materialize optimizer state for broadcast; "
                     "skipping (will sync after first step)")
 
+        from .moe.utils import is_moe_param
+
+        # state_dict keys params by flat index over param_groups; map the
+        # index back to the live parameter so MoE expert state (rank-local)
+        # can be skipped alongside its parameter.
+        flat_params = [p for group in optimizer.param_groups
+                       for p in group["params"]]
         state = optimizer.state_dict()["state"]
         tensors = []
         scalars = {}
         for pid, pstate in sorted(state.items()):
+            if (isinstance(pid, int) and pid < len(flat_params)
+                    and is_moe_param(flat_params[pid])):
+                continue
             for key, value in sorted(pstate.items()):
                 if isinstance(value, torch.Tensor):
                     tensors.append(value.data)

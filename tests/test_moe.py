@@ -87,6 +87,63 @@ def test_moe_train_top2():
     assert torch.equal(out[0][0], out[1][0])
 
 
+def _worker_moe_init_broadcast(rank, nprocs):
+    """Regression: the init state broadcast must NOT overwrite rank-local
+    expert parameters with rank 0's (engine broadcast used the
+    full state_dict; reference broadcasts bagua_build_params() which
+    excludes MoE params — bagua_distributed.py:172)."""
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+    from bagua_amd.parallel.moe import is_moe_param
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)  # per-rank expert init
+    model = MoEModel(k=1)
+    optimizer = torch.optim.Adam(model.parameters(), lr=1e-3)
+    # materialize per-rank optimizer state for the experts BEFORE wrapping,
+    # so the optimizer-state broadcast path is exercised too
+    out, l_aux = model(torch.randn(8, 8))
+    (out.sum() + 0.01 * l_aux).backward()
+    optimizer.step()
+    optimizer.zero_grad()
+    expert_state_before = {
+        id(p): {k: v.clone() for k, v in optimizer.state[p].items()
+                if isinstance(v, torch.Tensor)}
+        for p in model.parameters() if is_moe_param(p) and p in
+        optimizer.state}
+
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm())
+    del ddp
+
+    dense = torch.cat([p.detach().reshape(-1)
+                       for n, p in sorted(model.named_parameters())
+                       if not is_moe_param(p)])
+    expert = torch.cat([p.detach().reshape(-1)
+                        for n, p in sorted(model.named_parameters())
+                        if is_moe_param(p)])
+    # expert optimizer state must also survive the broadcast
+    state_ok = all(
+        torch.equal(optimizer.state[p][k], expert_state_before[id(p)][k])
+        for p in model.parameters()
+        if is_moe_param(p) and id(p) in expert_state_before
+        for k in expert_state_before[id(p)])
+    bagua_amd.deinit_process_group()
+    return dense, expert, state_ok
+
+
+def test_moe_init_broadcast_preserves_experts():
+    out = run_multi_process(2, _worker_moe_init_broadcast)
+    assert torch.equal(out[0][0], out[1][0]), "dense params not broadcast"
+    assert not torch.equal(out[0][1], out[1][1]), (
+        "init broadcast clobbered rank-local expert params")
+    assert out[0][2] and out[1][2], (
+        "init broadcast clobbered rank-local expert optimizer state")
+
+
 def _worker_moe_checkpoint(rank, nprocs, path):
     import bagua_amd
     from bagua_amd.checkpoint import load_checkpoint, save_checkpoint
