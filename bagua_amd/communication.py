@@ -463,10 +463,19 @@ class BaguaProcessGroup:
     become intra-node only.
     """
 
-    def __init__(self, ranks: List[int], stream, group_name: str):
+    def __init__(self, ranks: List[int], stream, group_name: str,
+                 dedicated: bool = False):
         self.ranks = list(ranks)
         self.stream = stream
         self.group_name = group_name
+        # dedicated groups own their torch process group (thread-safe vs
+        # the default group); the eager call below happens at group
+        # construction, which is a collective point on every rank —
+        # dist.new_group must never be first-called from a lazy path that
+        # only some ranks (or a background thread) reach.
+        self._comm_tag = group_name if dedicated else ""
+        if dedicated and dist.is_initialized():
+            _cached_torch_group(tuple(self.ranks), self._comm_tag)
         self._global_comm = None
         self._inter_comm = None
         self._intra_comm = None
@@ -494,7 +503,8 @@ class BaguaProcessGroup:
     def get_global_communicator(self) -> BaguaCommunicator:
         if self._global_comm is None:
             self._global_comm = _make_communicator(
-                self.group_name + "_global", self.ranks, self.stream)
+                self.group_name + "_global", self.ranks, self.stream,
+                self._comm_tag)
         return self._global_comm
 
     def get_inter_node_communicator(self) -> BaguaCommunicator:
@@ -505,22 +515,64 @@ class BaguaProcessGroup:
 
     def get_intra_node_communicator(self) -> BaguaCommunicator:
         if self._intra_comm is None:
+            # name must be unique PER NODE: every node's intra communicator
+            # has its own rank 0 writing an RCCL unique id to the store, so
+            # a shared name would collide across nodes (last-write-wins uid
+            # mixing => ncclCommInitRank hang).
+            my_node = self._rank_mappings()[env.get_rank()][0]
             self._intra_comm = _make_communicator(
-                self.group_name + "_intra", self._get_intra_ranks(), self.stream)
+                "{}_intra_n{}".format(self.group_name, my_node),
+                self._get_intra_ranks(), self.stream)
         return self._intra_comm
+
+    def ensure_native_communicators(self, hierarchical: bool = False):
+        """Eagerly construct the native RCCL communicator(s) at a
+        deterministic collective point (engine init), instead of lazily at
+        first use mid-backward. All ranks must call this in the same order.
+        No-op on CPU."""
+        if not _is_cuda_job():
+            return
+        self.get_global_communicator().ensure_native()
+        if hierarchical:
+            intra = self.get_intra_node_communicator()
+            if intra.nranks() < len(self.ranks):
+                intra.ensure_native()
+                if intra.rank_in_comm == 0:
+                    self.get_inter_node_communicator().ensure_native()
 
 
 @lru_cache(maxsize=None)
-def _cached_torch_group(ranks: tuple):
+def _cached_torch_group(ranks: tuple, tag: str = ""):
+    """One torch group per (ranks, tag).
+
+    tag="" communicators share a group per rank-set (engine comms are
+    driven from one thread, so sharing is safe). A non-empty tag forces a
+    DEDICATED torch group: two communicators over the same ranks used from
+    different threads (async model average's background loop vs the main
+    thread) must not share a gloo/RCCL channel — concurrent collectives on
+    one channel interleave across ranks and corrupt each other (observed
+    as gloo "received data size doesn't match" at world 8).
+    """
+    if tag:
+        # dedicated group: plain (globally-synchronized) creation; every
+        # rank calls this at the same collective point
+        return dist.new_group(list(ranks))
     if list(ranks) == list(range(dist.get_world_size())):
         return c10d._get_default_group()
-    return dist.new_group(list(ranks))
+    # Subset groups (intra-node, inter-node leaders) are created only by
+    # the ranks that reach them — node 0's ranks build [0..3] while node
+    # 1's build [4..7]. Counter-named new_group would collide those
+    # disjoint creations (observed hang at 2x4 pseudo-nodes);
+    # use_local_synchronization hashes the name from the rank set and
+    # returns None on non-members, making per-subset creation safe.
+    return dist.new_group(list(ranks), use_local_synchronization=True)
 
 
-def _make_communicator(name: str, ranks: List[int], stream) -> BaguaCommunicator:
+def _make_communicator(name: str, ranks: List[int], stream,
+                       tag: str = "") -> BaguaCommunicator:
     # torch group creation must be called by ALL ranks with the same list;
     # BaguaProcessGroup construction is collective, same as the reference.
-    torch_group = _cached_torch_group(tuple(ranks))
+    torch_group = _cached_torch_group(tuple(ranks), tag)
     return BaguaCommunicator(name, ranks, stream, torch_group)
 
 
@@ -554,8 +606,16 @@ def _get_default_group() -> BaguaProcessGroup:
 
 
 def new_group(ranks: Optional[List[int]] = None, stream=None,
-              group_name: Optional[str] = None) -> BaguaProcessGroup:
-    """Create a new process group (reference: communication.py:206-276)."""
+              group_name: Optional[str] = None,
+              dedicated: bool = False) -> BaguaProcessGroup:
+    """Create a new process group (reference: communication.py:206-276).
+
+    ``dedicated=True`` gives the group its own torch process group even
+    when its rank set matches an existing one — required when the group's
+    collectives run on a different host thread (async model average).
+    Collective: every rank must call with the same arguments in the same
+    order.
+    """
     _check_default_pg()
     if ranks is None:
         ranks = list(range(dist.get_world_size()))
@@ -564,7 +624,7 @@ def new_group(ranks: Optional[List[int]] = None, stream=None,
         stream = _new_comm_stream()
     if group_name is None:
         group_name = "group_" + "_".join(str(r) for r in ranks)
-    return BaguaProcessGroup(ranks, stream, group_name)
+    return BaguaProcessGroup(ranks, stream, group_name, dedicated=dedicated)
 
 
 def from_torch_group(group, stream=None) -> BaguaProcessGroup:

@@ -49,10 +49,15 @@ class AsyncModelAverageAlgorithmImpl(AlgorithmImpl):
         self.status = _AsyncInternalState.NEW
         self.future = None
 
-        # dedicated group + stream for the background loop
+        # dedicated group + stream for the background loop; ``dedicated``
+        # gives it its own torch/RCCL channel so the loop's collectives
+        # can never interleave with main-thread collectives on the default
+        # communicator (reference: async_model_average.py:72-82 also used
+        # a separate comm + stream)
         self.thread_group = new_group(
             process_group.ranks,
-            group_name=process_group.group_name + "_async")
+            group_name=process_group.group_name + "_async",
+            dedicated=True)
 
     # ------------------------------------------------------------------
     def tensors_to_buckets(self, tensors, do_flatten) -> List[BaguaBucket]:
@@ -133,6 +138,12 @@ class AsyncModelAverageAlgorithmImpl(AlgorithmImpl):
                 sync_interval_ms=self.sync_interval_ms,
                 group=self.thread_group)
             bucket._async_op = op
+        # build the dedicated RCCL communicator NOW, on the main thread at
+        # a point every rank reaches in the same order — constructing it
+        # lazily from the background loop's first allreduce would race the
+        # main thread's collectives on the default communicator
+        # (ncclCommInitRank is itself collective).
+        self.thread_group.ensure_native_communicators()
 
     # ------------------------------------------------------------------
     def _sync_device(self):

@@ -302,6 +302,13 @@ class BaguaDistributedDataParallel:
             groups, self.gradient_as_bucket_view)
         for bucket in self.bagua_buckets:
             self.bagua_algorithm.init_operations(self, bucket)
+        # construct every RCCL communicator the schedule will need NOW, at
+        # this collective point, rather than lazily inside the first
+        # backward (ncclCommInitRank is collective; deterministic ordering
+        # across ranks is what makes the 8-GPU first step land).
+        hierarchical = any(getattr(op, "hierarchical", False)
+                           for b in self.bagua_buckets for op in b.ops)
+        self.process_group.ensure_native_communicators(hierarchical)
         self.bagua_backend.register_ordered_buckets(self.bagua_buckets)
 
     # ------------------------------------------------------------------

@@ -31,6 +31,13 @@ def _worker(fn, rank, nprocs, port, args, queue):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     try:
+        # N procs x default-N intra-op threads thrashes the box at world 8
+        import torch
+
+        torch.set_num_threads(max(1, (os.cpu_count() or 1) // nprocs))
+    except Exception:
+        pass
+    try:
         result = fn(rank, nprocs, *args)
         queue.put((rank, "ok", pickle.dumps(result)))
     except Exception:
