@@ -838,6 +838,129 @@ int bagua_nccl_unique_id(char* out, size_t cap) {
   return (int)sizeof(id);
 }
 
+// --- full collective mirror for non-Python hosts (reference:
+// bagua-core-c/src/lib.rs:23-347 exposed the whole communicator surface;
+// dtype codes: 0=f32 1=f16 2=bf16 3=u8 4=i64) -------------------------
+
+static size_t c_esize(int dtype) {
+  switch (dtype) {
+    case 3: return 1;
+    case 1: case 2: return 2;
+    case 4: return 8;
+    default: return 4;
+  }
+}
+
+int bagua_comm_reduce_inplace(void* comm, void* ptr, size_t numel,
+                              int dtype, int dst, int op,
+                              uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  if (c->nranks() == 1) return 0;
+  return ncclReduce(ptr, ptr, numel, c_dtype(dtype), nccl_op(op), dst,
+                    c->raw(), (hipStream_t)stream) == ncclSuccess ? 0 : -1;
+}
+
+int bagua_comm_reduce_scatter_inplace(void* comm, void* ptr, size_t numel,
+                                      int dtype, int op, uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  if (c->nranks() == 1) return 0;
+  if (numel % c->nranks() != 0) return -2;
+  size_t chunk = numel / c->nranks();
+  char* own = (char*)ptr + c->rank() * chunk * c_esize(dtype);
+  return ncclReduceScatter(ptr, own, chunk, c_dtype(dtype), nccl_op(op),
+                           c->raw(), (hipStream_t)stream) == ncclSuccess
+             ? 0 : -1;
+}
+
+int bagua_comm_alltoall(void* comm, const void* send, void* recv,
+                        size_t numel, int dtype, uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  if (c->nranks() == 1) {
+    return hipMemcpyAsync(recv, send, numel * c_esize(dtype),
+                          hipMemcpyDeviceToDevice,
+                          (hipStream_t)stream) == hipSuccess ? 0 : -1;
+  }
+  if (numel % c->nranks() != 0) return -2;
+  return ncclAllToAll(send, recv, numel / c->nranks(), c_dtype(dtype),
+                      c->raw(), (hipStream_t)stream) == ncclSuccess
+             ? 0 : -1;
+}
+
+int bagua_comm_send(void* comm, const void* ptr, size_t numel, int dtype,
+                    int dst, uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  return ncclSend(ptr, numel, c_dtype(dtype), dst, c->raw(),
+                  (hipStream_t)stream) == ncclSuccess ? 0 : -1;
+}
+
+int bagua_comm_recv(void* comm, void* ptr, size_t numel, int dtype,
+                    int src, uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  return ncclRecv(ptr, numel, c_dtype(dtype), src, c->raw(),
+                  (hipStream_t)stream) == ncclSuccess ? 0 : -1;
+}
+
+int bagua_comm_gather(void* comm, const void* send, void* recv,
+                      size_t numel, int dtype, int dst, uintptr_t stream) {
+  // grouped p2p gather: every rank sends its chunk to dst
+  Communicator* c = (Communicator*)comm;
+  size_t es = c_esize(dtype);
+  if (ncclGroupStart() != ncclSuccess) return -1;
+  if (c->rank() == dst) {
+    for (int r = 0; r < c->nranks(); ++r) {
+      if (ncclRecv((char*)recv + (size_t)r * numel * es, numel,
+                   c_dtype(dtype), r, c->raw(),
+                   (hipStream_t)stream) != ncclSuccess)
+        return -1;
+    }
+  }
+  if (ncclSend(send, numel, c_dtype(dtype), dst, c->raw(),
+               (hipStream_t)stream) != ncclSuccess)
+    return -1;
+  return ncclGroupEnd() == ncclSuccess ? 0 : -1;
+}
+
+int bagua_comm_scatter(void* comm, const void* send, void* recv,
+                       size_t numel, int dtype, int src, uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  size_t es = c_esize(dtype);
+  if (ncclGroupStart() != ncclSuccess) return -1;
+  if (c->rank() == src) {
+    for (int r = 0; r < c->nranks(); ++r) {
+      if (ncclSend((const char*)send + (size_t)r * numel * es, numel,
+                   c_dtype(dtype), r, c->raw(),
+                   (hipStream_t)stream) != ncclSuccess)
+        return -1;
+    }
+  }
+  if (ncclRecv(recv, numel, c_dtype(dtype), src, c->raw(),
+               (hipStream_t)stream) != ncclSuccess)
+    return -1;
+  return ncclGroupEnd() == ncclSuccess ? 0 : -1;
+}
+
+int bagua_comm_group_start(void) {
+  return ncclGroupStart() == ncclSuccess ? 0 : -1;
+}
+
+int bagua_comm_group_end(void) {
+  return ncclGroupEnd() == ncclSuccess ? 0 : -1;
+}
+
+// barrier = 1-element allreduce (reference: communication.py:1377-1401),
+// followed by a stream sync so the host really is past it
+int bagua_comm_barrier(void* comm, void* scratch_one_elem,
+                       uintptr_t stream) {
+  Communicator* c = (Communicator*)comm;
+  if (c->nranks() > 1) {
+    if (ncclAllReduce(scratch_one_elem, scratch_one_elem, 1, ncclFloat32,
+                      ncclSum, c->raw(),
+                      (hipStream_t)stream) != ncclSuccess)
+      return -1;
+  }
+  return hipStreamSynchronize((hipStream_t)stream) == hipSuccess ? 0 : -1;
+}
+
 }  // extern "C"
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

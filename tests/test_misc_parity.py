@@ -263,7 +263,15 @@ def test_c_abi_symbols_present():
     for sym in ("bagua_comm_create", "bagua_comm_destroy",
                 "bagua_comm_rank", "bagua_comm_nranks", "bagua_comm_abort",
                 "bagua_comm_allreduce_inplace", "bagua_comm_broadcast",
-                "bagua_comm_allgather_inplace", "bagua_nccl_unique_id"):
+                "bagua_comm_allgather_inplace", "bagua_nccl_unique_id",
+                # full collective mirror (reference:
+                # bagua-core-c/src/lib.rs:23-347)
+                "bagua_comm_reduce_inplace",
+                "bagua_comm_reduce_scatter_inplace",
+                "bagua_comm_alltoall", "bagua_comm_send", "bagua_comm_recv",
+                "bagua_comm_gather", "bagua_comm_scatter",
+                "bagua_comm_group_start", "bagua_comm_group_end",
+                "bagua_comm_barrier"):
         assert hasattr(lib, sym), "missing C ABI symbol %s" % sym
 
 
