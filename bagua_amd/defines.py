@@ -6,7 +6,7 @@ autotuned hyperparameter space is sized for xGMI bucket fusion.
 """
 
 import enum
-from typing import Dict, List
+from typing import Dict, List, Optional
 
 from pydantic import BaseModel
 
@@ -41,7 +41,9 @@ class BaguaHyperparameter(BaseModel):
 
     buckets: List[List[TensorDeclaration]] = []
     bucket_size: int = 32 * 1024 * 1024
-    is_hierarchical_reduce: bool = False
+    # None = the tuner is not searching this dimension (single node);
+    # the engine then leaves the algorithm's own flag alone
+    is_hierarchical_reduce: Optional[bool] = None
 
     def update(self, param_dict: Dict) -> "BaguaHyperparameter":
         tmp = self.dict()

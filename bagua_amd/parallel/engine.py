@@ -415,6 +415,16 @@ class BaguaDistributedDataParallel:
         self._autotune_completed = bool(rsp.get("is_autotune_completed",
                                                 False))
         self._current_hp = hp
+        # consume the hierarchical dimension (the reference searched it
+        # but nothing read it, autotune_task_manager.py:107-113): flip the
+        # algorithm's flag so the re-bucketed ops take the intra-reduce /
+        # inter-op / intra-bcast path when the tuner asks for it. None
+        # means the tuner is not searching it (single node) — leave the
+        # user's flag alone.
+        if (hp.is_hierarchical_reduce is not None
+                and hasattr(self.bagua_algorithm, "hierarchical")):
+            self.bagua_algorithm.hierarchical = bool(
+                hp.is_hierarchical_reduce)
         return hp
 
     def _bagua_autotune_step(self):

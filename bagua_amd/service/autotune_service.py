@@ -18,6 +18,7 @@ work between requests).
 
 import json
 import logging
+import os
 import threading
 import time
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
@@ -37,13 +38,15 @@ class AutotuneServiceState:
     def __init__(self, world_size: int, autotune_level: int = 1,
                  max_samples: int = 60, sampling_confidence_time_s: float = 5.0,
                  warmup_time_s: float = 30.0,
-                 default_bucket_size: int = 32 * 1024 * 1024):
+                 default_bucket_size: int = 32 * 1024 * 1024,
+                 nnodes: int = 1):
         self.world_size = world_size
         self.autotune_level = autotune_level
         self.max_samples = max_samples
         self.sampling_confidence_time_s = sampling_confidence_time_s
         self.warmup_time_s = warmup_time_s
         self.default_bucket_size = default_bucket_size
+        self.nnodes = nnodes
 
         self.lock = threading.Lock()
         self.managers: Dict[str, AutotuneTaskManager] = {}
@@ -55,11 +58,18 @@ class AutotuneServiceState:
         self.start_time = time.time()
         # metric reports: model -> {rank: (iter, speed)}
         self.metrics: Dict[str, Dict[int, tuple]] = {}
+        # train_iter at which the current hp was handed out: a speed
+        # sample only scores the current hp if it was MEASURED under it
+        # (reference gated per-iteration with all ranks synced,
+        # autotune_service.py:78-272; mixed-iteration samples would score
+        # a hyperparameter with stale speeds)
+        self.hp_installed_iter: Dict[str, int] = {}
 
     def manager(self, model_name: str) -> AutotuneTaskManager:
         if model_name not in self.managers:
             self.managers[model_name] = AutotuneTaskManager(
-                model_name, env.is_output_autotune_log())
+                model_name, env.is_output_autotune_log(),
+                search_hierarchical=self.nnodes > 1)
         return self.managers[model_name]
 
 
@@ -158,9 +168,13 @@ class _Handler(BaseHTTPRequestHandler):
         if time.time() - st.start_time < st.warmup_time_s:
             return {"recommended_hyperparameters": hp.dict(),
                     "is_autotune_completed": False}
-        # wait until every rank reported a speed sample for this hp
+        # wait until every rank reported a speed sample MEASURED UNDER the
+        # current hp (report train_iter newer than the hp's install
+        # iteration) — the all-ranks-synced-per-iteration gate
         reports = st.metrics.get(model, {})
-        if len(reports) < st.world_size:
+        installed = st.hp_installed_iter.get(model, 0)
+        fresh = {r: v for r, v in reports.items() if v[0] > installed}
+        if len(fresh) < st.world_size:
             return {"recommended_hyperparameters": hp.dict(),
                     "is_autotune_completed": False}
         # confidence gate: at most one proposal per confidence window
@@ -170,11 +184,12 @@ class _Handler(BaseHTTPRequestHandler):
             return {"recommended_hyperparameters": hp.dict(),
                     "is_autotune_completed": False}
 
-        score = sum(v[1] for v in reports.values()) / len(reports)
+        score = sum(v[1] for v in fresh.values()) / len(fresh)
         mgr.record(train_iter, hp, score)
         st.sample_count[model] = st.sample_count.get(model, 0) + 1
         st.metrics[model] = {}
         st.last_proposal_time[model] = now
+        st.hp_installed_iter[model] = train_iter
 
         if st.sample_count[model] >= st.max_samples:
             st.completed[model] = True
@@ -202,6 +217,10 @@ class AutotuneServer:
                 env.get_autotune_sampling_confidence_time_s()),
             warmup_time_s=env.get_autotune_warmup_time_s(),
             default_bucket_size=env.get_default_bucket_size(),
+            # LOCAL_WORLD_SIZE absent => launched outside torchrun;
+            # assume a single node (searching hierarchical there is noise)
+            nnodes=(max(1, world_size // max(1, env.get_local_size()))
+                    if "LOCAL_WORLD_SIZE" in os.environ else 1),
         )
         handler = type("BoundHandler", (_Handler,), {"state": state})
         self.state = state

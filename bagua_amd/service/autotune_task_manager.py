@@ -46,16 +46,24 @@ def split_bucket_by_bucket_size(
 
 
 class AutotuneTaskManager:
-    def __init__(self, model_name: str, is_output_log: bool = False):
+    def __init__(self, model_name: str, is_output_log: bool = False,
+                 search_hierarchical: bool = False):
         self.model_name = model_name
         self.records: List[Tuple[int, BaguaHyperparameter, float]] = []
         self.tensor_list: List[TensorDeclaration] = []
         # tensor_name -> order index derived from telemetry spans
         self.tensor_order: Dict[str, int] = {}
-        self.optimizer = BayesianOptimizer({
-            "bucket_size_2p": IntParam(25, (20, 31)),
-            "is_hierarchical_reduce": BoolParam(False),
-        })
+        # hierarchical reduce only changes execution on MULTI-node runs
+        # (one node: intra == global and the ops take the flat path), so
+        # searching it on a single node injects pure noise into the
+        # hill-climb — the reference searched-but-never-consumed it
+        # (autotune_task_manager.py:48-59); here it is searched only when
+        # it can matter and IS consumed (engine._ask_hyperparameters).
+        space = {"bucket_size_2p": IntParam(25, (20, 31))}
+        if search_hierarchical:
+            space["is_hierarchical_reduce"] = BoolParam(False)
+        self.search_hierarchical = search_hierarchical
+        self.optimizer = BayesianOptimizer(space)
         self.sampling_start: Optional[float] = None
         self._log_writer = None
         if is_output_log:
@@ -102,15 +110,20 @@ class AutotuneTaskManager:
 
     def tell_and_ask(self, prev_hp: BaguaHyperparameter,
                      prev_score: float) -> BaguaHyperparameter:
-        self.optimizer.tell({
+        point = {
             "bucket_size_2p": max(prev_hp.bucket_size, 1).bit_length() - 1,
-            "is_hierarchical_reduce": prev_hp.is_hierarchical_reduce,
-        }, prev_score)
+        }
+        if self.search_hierarchical:
+            point["is_hierarchical_reduce"] = bool(
+                prev_hp.is_hierarchical_reduce)
+        self.optimizer.tell(point, prev_score)
         proposal = self.optimizer.ask()
         bucket_size = 1 << int(proposal["bucket_size_2p"])
+        hier = (bool(proposal["is_hierarchical_reduce"])
+                if self.search_hierarchical else None)
         hp = BaguaHyperparameter(
             bucket_size=bucket_size,
-            is_hierarchical_reduce=bool(proposal["is_hierarchical_reduce"]),
+            is_hierarchical_reduce=hier,
             buckets=split_bucket_by_bucket_size(
                 self.ordered_tensor_list(), bucket_size),
         )

@@ -86,3 +86,64 @@ def test_split_bucket_by_bucket_size():
     # a+b fit one f32 bucket; c splits on dtype; d overflows
     assert [[t.name for t in b] for b in buckets] == [
         ["a", "b"], ["c"], ["d"]]
+
+
+def test_stale_iteration_samples_do_not_score(server):
+    """All-ranks-synced-per-hp gate: a speed sample reported for an
+    iteration at or before the current hp's install point must not score
+    it (reference gated per-iteration with all ranks synced,
+    autotune_service.py:78-272)."""
+    client = AutotuneClient("127.0.0.1", server.port)
+    client.register_tensors("m2", _decls())
+    st = server.state
+
+    # both ranks report at iter 5, then a proposal happens at iter 5
+    for rank in range(2):
+        client.report_metrics("m2", rank, 5, {}, 100.0)
+    client.ask_hyperparameters("m2", 0, 5)
+    assert st.sample_count["m2"] == 1
+    installed = st.hp_installed_iter["m2"]
+    assert installed == 5
+
+    # rank 0 reports a FRESH sample, rank 1 a stale one (iter <= installed)
+    client.report_metrics("m2", 0, 7, {}, 100.0)
+    client.report_metrics("m2", 1, installed, {}, 100.0)
+    client.ask_hyperparameters("m2", 0, 7)
+    assert st.sample_count["m2"] == 1, "stale sample scored the hp"
+
+    # once rank 1 is fresh too the proposal goes through
+    client.report_metrics("m2", 1, 7, {}, 100.0)
+    client.ask_hyperparameters("m2", 0, 7)
+    assert st.sample_count["m2"] == 2
+
+
+def test_single_node_does_not_search_hierarchical(server):
+    """One node => the hierarchical dimension is pure noise; it must not
+    be in the search space and proposals must carry None so the engine
+    leaves the user's algorithm flag alone."""
+    client = AutotuneClient("127.0.0.1", server.port)
+    client.register_tensors("m3", _decls())
+    st = server.state
+    mgr = st.manager("m3")
+    assert not mgr.search_hierarchical
+    for it in range(1, 8):
+        for rank in range(2):
+            client.report_metrics("m3", rank, it, {}, 100.0)
+        rsp = client.ask_hyperparameters("m3", 0, it)
+        hp = rsp["recommended_hyperparameters"]
+        assert hp.get("is_hierarchical_reduce") is None
+
+
+def test_multi_node_searches_hierarchical():
+    """nnodes>1 puts is_hierarchical_reduce back into the search space."""
+    from bagua_amd.service.autotune_task_manager import AutotuneTaskManager
+
+    mgr = AutotuneTaskManager("m", search_hierarchical=True)
+    assert mgr.search_hierarchical
+    hp = BaguaHyperparameter(bucket_size=1 << 25)
+    seen = set()
+    for i in range(12):
+        hp = mgr.tell_and_ask(hp, 100.0 + i)
+        assert hp.is_hierarchical_reduce in (True, False)
+        seen.add(hp.is_hierarchical_reduce)
+    assert len(seen) == 2, "hierarchical dimension never explored"
