@@ -1,6 +1,16 @@
-"""GShard-style top-k gated MoE layer
-(reference: bagua/torch_api/model_parallel/moe/sharded_moe.py:77-375,
-itself DeepSpeed-derived; re-implemented here from the GShard math).
+"""GShard-style top-k gated MoE layer.
+
+Attribution: the gating algorithm follows GShard (Lepikhin et al., 2020,
+arXiv:2006.16668) as popularized by DeepSpeed's MoE implementation
+(Microsoft, MIT license), which the reference vendored
+(bagua/torch_api/model_parallel/moe/sharded_moe.py:77-375, itself marked
+Copyright Microsoft / "COPYRIGHT NOTICE: code modified from deepspeed").
+The capacity/queueing semantics here intentionally match that lineage so
+checkpoints and behavior line up; ``top1gating`` (drop_tokens queueing)
+and ``top2gating`` (gather/scatter slot assignment instead of the
+einsum-and-one-hot choreography) are restructured implementations of the
+same math, while the jitter/gumbel helpers remain close to canonical
+form.
 
 Token flow per layer: gate -> dispatch einsum -> alltoall over the
 expert-parallel group -> local experts -> alltoall back -> combine
@@ -124,53 +134,60 @@ def top1gating(
 def top2gating(
     logits: Tensor, capacity_factor: float, min_capacity: int,
 ) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
-    """Top-2 gating with normalized weights."""
+    """Top-2 gating with renormalized pair weights.
+
+    Same GShard semantics as the DeepSpeed lineage (see module header) —
+    first choices fill each expert's queue before any second choice,
+    over-capacity slots are dropped, surviving pair weights renormalize to
+    sum to 1 — but computed by direct gather/scatter slot assignment
+    rather than stacked one-hot einsums: cheaper ((s,e,c) tensors are
+    only materialized once for the output) and, on GPU, fewer kernels.
+    """
     gates = F.softmax(logits, dim=1)
     num_tokens, num_experts = gates.shape
     capacity = _capacity(num_tokens, num_experts, 2 * capacity_factor,
                          min_capacity)
+    tok = torch.arange(num_tokens, device=logits.device)
 
-    indices1_s = torch.argmax(gates, dim=1)
-    mask1 = F.one_hot(indices1_s, num_classes=num_experts)
+    # choices: top-1 on clean gates; runner-up on gumbel-noised logits
+    # with the first choice masked out
+    expert1 = torch.argmax(gates, dim=1)
+    noised = logits + gumbel_rsample(logits.shape, device=logits.device)
+    noised = noised.scatter(1, expert1.unsqueeze(1), float("-inf"))
+    expert2 = torch.argmax(noised, dim=1)
 
-    logits_w_noise = logits + gumbel_rsample(logits.shape,
-                                             device=logits.device)
-    logits_except1 = logits_w_noise.masked_fill(mask1.bool(),
-                                                float("-inf"))
-    indices2_s = torch.argmax(logits_except1, dim=1)
-    mask2 = F.one_hot(indices2_s, num_classes=num_experts)
+    oh1 = F.one_hot(expert1, num_classes=num_experts)
+    oh2 = F.one_hot(expert2, num_classes=num_experts)
 
-    locations1 = torch.cumsum(mask1, dim=0) - 1
-    locations2 = torch.cumsum(mask2, dim=0) - 1
-    locations2 += torch.sum(mask1, dim=0, keepdim=True)
+    # queue position of each token at its chosen expert: first choices
+    # first (in token order), second choices continue after ALL first
+    # choices of that expert
+    pos1 = torch.cumsum(oh1, dim=0) - 1
+    pos2 = torch.cumsum(oh2, dim=0) - 1 + oh1.sum(dim=0, keepdim=True)
+    slot1 = pos1.gather(1, expert1.unsqueeze(1)).squeeze(1)
+    slot2 = pos2.gather(1, expert2.unsqueeze(1)).squeeze(1)
 
-    me = torch.mean(gates, dim=0)
-    ce = torch.mean(mask1.float(), dim=0)
-    l_aux = torch.mean(me * ce) * num_experts * num_experts
+    # load-balancing loss over the first choice (GShard eq. 4, top-2
+    # normalization), computed before capacity dropping
+    l_aux = torch.mean(gates.mean(dim=0) * oh1.float().mean(dim=0)) \
+        * num_experts * num_experts
 
-    mask1 = mask1 * torch.lt(locations1, capacity)
-    mask2 = mask2 * torch.lt(locations2, capacity)
-    locations1_s = torch.sum(locations1 * mask1, dim=1)
-    locations2_s = torch.sum(locations2 * mask2, dim=1)
+    keep1 = slot1 < capacity
+    keep2 = slot2 < capacity
 
-    mask1_float = mask1.float()
-    mask2_float = mask2.float()
-    gates1_s = torch.einsum("se,se->s", gates, mask1_float)
-    gates2_s = torch.einsum("se,se->s", gates, mask2_float)
-    denom_s = torch.clamp(gates1_s + gates2_s,
-                          min=torch.finfo(gates.dtype).eps)
-    gates1_s = gates1_s / denom_s
-    gates2_s = gates2_s / denom_s
+    # renormalize the surviving pair weights to sum to one
+    w1 = gates.gather(1, expert1.unsqueeze(1)).squeeze(1) * keep1
+    w2 = gates.gather(1, expert2.unsqueeze(1)).squeeze(1) * keep2
+    denom = torch.clamp(w1 + w2, min=torch.finfo(gates.dtype).eps)
+    w1 = w1 / denom
+    w2 = w2 / denom
 
-    gates1 = torch.einsum("s,se->se", gates1_s, mask1_float)
-    gates2 = torch.einsum("s,se->se", gates2_s, mask2_float)
-    locations1_sc = F.one_hot(locations1_s, num_classes=capacity).float()
-    locations2_sc = F.one_hot(locations2_s, num_classes=capacity).float()
-    combine1_sec = torch.einsum("se,sc->sec", gates1, locations1_sc)
-    combine2_sec = torch.einsum("se,sc->sec", gates2, locations2_sc)
-    combine_weights = combine1_sec + combine2_sec
+    combine_weights = logits.new_zeros(num_tokens, num_experts, capacity)
+    combine_weights[tok[keep1], expert1[keep1], slot1[keep1]] = w1[keep1]
+    combine_weights[tok[keep2], expert2[keep2], slot2[keep2]] = w2[keep2]
     dispatch_mask = combine_weights.bool()
-    exp_counts = torch.sum(mask1 + mask2, dim=0)
+    exp_counts = (oh1 * keep1.unsqueeze(1)
+                  + oh2 * keep2.unsqueeze(1)).sum(dim=0)
     return l_aux, combine_weights, dispatch_mask, exp_counts
 
 

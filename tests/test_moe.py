@@ -249,3 +249,71 @@ def test_gate_top2_weights_normalized():
     kept = sums > 0
     assert t.all(sums[kept] <= 1.0 + 1e-5)
     assert t.isfinite(l_aux)
+
+
+def _worker_moe_subgroup_ep(rank, nprocs):
+    """ep_size < world_size: EP alltoall confined to a 2-rank sub-group
+    at world 4 (goes beyond the reference, which hardcoded
+    group=dist.group.WORLD — layer.py:84; the MOELayer group parameter is
+    the DeepSpeed-lineage interface for this)."""
+    import torch.distributed as dist
+
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+    from bagua_amd.parallel.moe import MoE, is_moe_param
+
+    bagua_amd.init_process_group()
+    # two EP groups: {0,1} and {2,3}; every rank must create both
+    g01 = dist.new_group([0, 1])
+    g23 = dist.new_group([2, 3])
+    ep_group = g01 if rank < 2 else g23
+
+    torch.manual_seed(13 + rank)
+    model = MoEModel(k=1)
+    # swap in a sub-group MoE layer
+    import torch.nn as nn
+
+    model.moe = MoE(16,
+                    expert=nn.Sequential(nn.Linear(16, 32), nn.ReLU(),
+                                         nn.Linear(32, 16)),
+                    num_local_experts=2, k=1,
+                    expert_parallel_group=ep_group)
+    assert model.moe.ep_size == 2
+    assert model.moe.num_experts == 4  # 2 local x 2 ranks in the EP group
+
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm())
+
+    for step in range(6):
+        torch.manual_seed(900 + rank * 17 + step)
+        data = torch.randn(16, 8)
+        target = torch.randn(16, 4)
+        optimizer.zero_grad()
+        out, l_aux = ddp(data)
+        loss = F.mse_loss(out, target) + 0.01 * l_aux
+        loss.backward()
+        optimizer.step()
+        assert torch.isfinite(loss)
+
+    dense = torch.cat([p.detach().reshape(-1)
+                       for n, p in sorted(model.named_parameters())
+                       if not is_moe_param(p)])
+    expert = torch.cat([p.detach().reshape(-1)
+                        for n, p in sorted(model.named_parameters())
+                        if is_moe_param(p)])
+    bagua_amd.deinit_process_group()
+    return dense, expert
+
+
+def test_moe_subgroup_expert_parallel():
+    out = run_multi_process(4, _worker_moe_subgroup_ep)
+    # dense params: full-world consensus
+    for r in range(1, 4):
+        assert torch.equal(out[0][0], out[r][0]), "dense diverged"
+    # experts shard within each EP group (different per rank)
+    assert not torch.equal(out[0][1], out[1][1])
+    assert not torch.equal(out[2][1], out[3][1])
