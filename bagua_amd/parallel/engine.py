@@ -190,6 +190,17 @@ class BaguaDistributedDataParallel:
             try:
                 optimizer.step()
                 optimizer.zero_grad()
+                # the zero-grad materialization step must be invisible to
+                # step-counter-sensitive optimizers (QAdam freezes its
+                # second moment at state["step"]==warmup_steps; torch
+                # Adam's bias correction reads it too) — rewind the
+                # counters the fake step advanced
+                for pstate in optimizer.state.values():
+                    step = pstate.get("step")
+                    if isinstance(step, int):
+                        pstate["step"] = 0
+                    elif isinstance(step, torch.Tensor):
+                        step.zero_()
             except Exception:
                 logger.debug(
                     "cannot materialize optimizer state for broadcast; "

@@ -70,12 +70,15 @@ def _simulate(nprocs):
             for m in models]
 
     # engine init: optimizer state materialization with zero grads
-    # (engine._bagua_broadcast_optimizer_state), consuming step 1
+    # (engine._bagua_broadcast_optimizer_state) — the engine rewinds the
+    # step counters afterwards so the fake step is invisible
     for m, o in zip(models, opts):
         for p in m.parameters():
             p.grad = torch.zeros_like(p)
         o.step()
         o.zero_grad()
+        for p in m.parameters():
+            o.state[p]["step"] = 0
 
     # registration order: reverse of build-params order, matching the
     # QAdam impl (sorted by _q_adam_idx)
