@@ -160,15 +160,19 @@ def _on_comm_stream(comm, fn):
     """Run local device math on the comm stream so it is ordered with the
     collectives without host syncs."""
     if comm.stream is not None and torch.cuda.is_available():
+        from .communication import _event_get, _event_put
+
         curr = torch.cuda.current_stream()
-        ev = torch.cuda.Event()
+        ev = _event_get()
         ev.record(curr)
         comm.stream.wait_event(ev)
+        _event_put(ev)
         with torch.cuda.stream(comm.stream):
             out = fn()
-        done = torch.cuda.Event()
+        done = _event_get()
         done.record(comm.stream)
         curr.wait_event(done)
+        _event_put(done)
         return out
     return fn()
 

@@ -161,6 +161,16 @@ class Communicator {
     }
   }
 
+  // graceful teardown from a known-good thread (deinit_process_group);
+  // relying on GC to run ~Communicator from an arbitrary thread leaks
+  // RCCL state in long-lived multi-model processes
+  void destroy() {
+    if (comm_) {
+      ncclCommDestroy(comm_);
+      comm_ = nullptr;
+    }
+  }
+
   ncclComm_t raw() const { return comm_; }
 
   void group_start() { NCCL_CHECK(ncclGroupStart()); }
@@ -598,10 +608,15 @@ class BucketExecutor {
         if (duration_cast<seconds>(now - p.second).count() > 300) {
           fprintf(stderr,
                   "[bagua_amd] FATAL: a scheduled communication has been "
-                  "running for >300s; aborting communicator\n");
+                  "running for >300s; aborting ALL communicators\n");
           fflush(stderr);
           lk.unlock();
+          // in hierarchical mode the wedged collective can live on ANY of
+          // the three communicators — abort them all so no intra/inter
+          // collective keeps the ranks pinned
           global_->abort();
+          if (intra_) intra_->abort();
+          if (inter_) inter_->abort();
           lk.lock();
           break;
         }
@@ -971,6 +986,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("rank", &Communicator::rank)
       .def("nranks", &Communicator::nranks)
       .def("abort", &Communicator::abort)
+      .def("destroy", &Communicator::destroy)
       .def("group_start", &Communicator::group_start)
       .def("group_end", &Communicator::group_end)
       .def("allreduce_inplace", &Communicator::allreduce_inplace)
