@@ -96,8 +96,38 @@ class DistributedDataParallel_V1_9_0(Module):
     def bagua_buckets(self):
         return self.inner.bagua_buckets
 
+    def _sync_buffers(self):
+        # torch-DDP semantics: with broadcast_buffers=True, module buffers
+        # (BN running stats etc.) are re-broadcast from rank 0 before each
+        # authoritative forward so ranks can never drift
+        # (test_c10d_common checks this behavior class).
+        from ..communication import broadcast_coalesced
+
+        bufs = [b.data for b in self.module.buffers() if b.numel() > 0]
+        if bufs:
+            comm = self.inner.process_group.get_global_communicator()
+            broadcast_coalesced(bufs, src=0, comm=comm)
+
     def forward(self, *inputs, **kwargs):
+        if (self.broadcast_buffers and self.module.training
+                and self.inner.require_backward_grad_sync
+                and torch.distributed.is_initialized()
+                and torch.distributed.get_world_size() > 1):
+            self._sync_buffers()
         return self.module(*inputs, **kwargs)
+
+    def register_comm_hook(self, state, hook):
+        """torch DDP comm hooks rewire c10d bucket allreduce; bagua's
+        algorithm abstraction IS that layer — use a custom Algorithm
+        instead (parity with the reference, which also rejected hooks)."""
+        raise NotImplementedError(
+            "bagua DistributedDataParallel does not support "
+            "register_comm_hook; implement a bagua Algorithm instead")
+
+    def _register_builtin_comm_hook(self, comm_hook_type):
+        raise NotImplementedError(
+            "bagua DistributedDataParallel does not support "
+            "register_comm_hook; implement a bagua Algorithm instead")
 
     @contextmanager
     def no_sync(self):

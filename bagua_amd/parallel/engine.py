@@ -115,6 +115,12 @@ class BaguaDistributedDataParallel:
             bucket_size=env.get_default_bucket_size())
         self._bagua_autograd_hook_handles = []
 
+        # validate the parameter set (sparse rejection etc.) BEFORE the
+        # state broadcast touches any tensor — a sparse parameter must
+        # fail with the explicit ValueError, not a reshape crash inside
+        # broadcast_coalesced (reference rejects at ctor,
+        # bagua_distributed.py:155-212)
+        self.bagua_build_params()
         self._install_forward_pre_hooks()
         self._bagua_init_algorithm()
 
