@@ -65,6 +65,11 @@ class BaguaBackend:
         self.telemetry_enabled = env.get_autotune_level() > 0
         self.telemetry_spans: List[dict] = []
         self._native_exec = None  # C++ BucketExecutor (GPU)
+        # host-side dispatch cost accounting (python-executor overhead
+        # evidence): ns spent inside _execute and how
+        # many bucket dispatches it covers
+        self.exec_host_ns = 0
+        self.exec_dispatches = 0
 
     # ------------------------------------------------------------------
     def register_ordered_buckets(self, buckets: List[BaguaBucket]):
@@ -169,6 +174,7 @@ class BaguaBackend:
             self._queue_idx = 0
 
     def _execute(self, bucket: BaguaBucket):
+        t0 = time.perf_counter_ns()
         if _ROCTX:
             # shows up as a marker range in rocprofv3 --marker-trace
             torch.cuda.nvtx.range_push("bagua_bucket:%s" % bucket.name)
@@ -177,6 +183,8 @@ class BaguaBackend:
         finally:
             if _ROCTX:
                 torch.cuda.nvtx.range_pop()
+            self.exec_host_ns += time.perf_counter_ns() - t0
+            self.exec_dispatches += 1
 
     def _execute_inner(self, bucket: BaguaBucket):
         if getattr(bucket, "_native_idx", None) is not None:
