@@ -390,7 +390,27 @@ class BaguaCommunicator:
 
         self._run(native, fallback, [send_tensor, recv_tensor])
 
+    def _p2p_alltoall_inplace(self, tensor) -> bool:
+        """Opt-in direct xGMI path (BAGUA_P2P_ALLTOALL=1): one-hop pull
+        alltoall through IPC-exported buffers instead of ncclAllToAll.
+        Returns False when not applicable (CPU, world 1, disabled)."""
+        from .ops import p2p
+
+        if (not p2p.enabled() or self.nranks() <= 1
+                or not _is_cuda_job() or not tensor.is_cuda):
+            return False
+        impl = p2p.get_for_communicator(
+            self, tensor.numel() * tensor.element_size())
+
+        def native():
+            impl.alltoall(tensor, tensor)
+
+        self._run(native, lambda: None, [tensor])
+        return True
+
     def alltoall_inplace(self, tensor):
+        if self._p2p_alltoall_inplace(tensor):
+            return
         def native():
             self._native.alltoall_inplace(tensor)
 

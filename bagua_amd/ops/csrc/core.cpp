@@ -637,6 +637,144 @@ class BucketExecutor {
 };
 
 // ---------------------------------------------------------------------------
+// Direct one-hop p2p alltoall over xGMI (SURVEY hard part #4 prototype).
+//
+// The 8-GPU MI355X node is fully connected (7 xGMI links per GPU), so
+// the natural alltoall is each rank PULLING its chunk from every peer's
+// exported buffer in one hop — no ring, no proxy. Buffers and a
+// fine-grained flag array are exported with hipIpcGetMemHandle and
+// opened by peers; rounds are ordered by a device-side sequence-number
+// barrier (kernels.hip p2p_barrier_kernel) because re-recorded IPC
+// events race across processes. The pulls are plain stream-ordered D2D
+// copies (SDMA engines move them over the direct link).
+//
+// Opt-in (BAGUA_P2P_ALLTOALL=1 wires it into the scattergather/ByteGrad
+// chunk exchange); default stays on ncclAllToAll.
+// ---------------------------------------------------------------------------
+
+extern "C" void bagua_p2p_barrier_launch(void* peer_flags_dev_array,
+                                         void* my_flags, int rank,
+                                         int nranks,
+                                         unsigned long long seq,
+                                         hipStream_t stream);
+
+class P2PAlltoAll {
+ public:
+  P2PAlltoAll(int rank, int nranks, uintptr_t stream, int64_t capacity)
+      : rank_(rank), nranks_(nranks), stream_((hipStream_t)stream),
+        capacity_(capacity) {
+    TORCH_CHECK(nranks_ >= 1 && nranks_ <= 64, "p2p alltoall: 1..64 ranks");
+    HIP_CHECK(hipMalloc(&send_, capacity_));
+    HIP_CHECK(hipMalloc(&recv_, capacity_));
+    // flags must be FINE-GRAINED for cross-device visibility of the
+    // barrier stores over xGMI
+    HIP_CHECK(hipExtMallocWithFlags(
+        (void**)&flags_, nranks_ * sizeof(unsigned long long),
+        hipDeviceMallocFinegrained));
+    HIP_CHECK(hipMemset((void*)flags_, 0,
+                        nranks_ * sizeof(unsigned long long)));
+    HIP_CHECK(hipMalloc(&d_peer_flags_, nranks_ * sizeof(void*)));
+    peer_send_.assign(nranks_, nullptr);
+    peer_flags_.assign(nranks_, nullptr);
+  }
+
+  ~P2PAlltoAll() {
+    for (int p = 0; p < nranks_; ++p) {
+      if (p != rank_ && peer_send_[p]) hipIpcCloseMemHandle(peer_send_[p]);
+      if (p != rank_ && peer_flags_[p])
+        hipIpcCloseMemHandle((void*)peer_flags_[p]);
+    }
+    hipFree(send_);
+    hipFree(recv_);
+    hipFree((void*)flags_);
+    hipFree(d_peer_flags_);
+  }
+
+  // serialized (send handle || flags handle) for the store exchange
+  py::bytes handles() {
+    hipIpcMemHandle_t h[2];
+    HIP_CHECK(hipIpcGetMemHandle(&h[0], send_));
+    HIP_CHECK(hipIpcGetMemHandle(&h[1], (void*)flags_));
+    return py::bytes((const char*)h, sizeof(h));
+  }
+
+  void connect(const std::vector<std::string>& peer_handles) {
+    TORCH_CHECK((int)peer_handles.size() == nranks_,
+                "need one handle blob per rank");
+    for (int p = 0; p < nranks_; ++p) {
+      if (p == rank_) {
+        peer_send_[p] = send_;
+        peer_flags_[p] = flags_;
+        continue;
+      }
+      TORCH_CHECK(peer_handles[p].size() == 2 * sizeof(hipIpcMemHandle_t),
+                  "bad handle blob size");
+      hipIpcMemHandle_t h[2];
+      std::memcpy(h, peer_handles[p].data(), sizeof(h));
+      void* ps = nullptr;
+      void* pf = nullptr;
+      HIP_CHECK(hipIpcOpenMemHandle(&ps, h[0],
+                                    hipIpcMemLazyEnablePeerAccess));
+      HIP_CHECK(hipIpcOpenMemHandle(&pf, h[1],
+                                    hipIpcMemLazyEnablePeerAccess));
+      peer_send_[p] = ps;
+      peer_flags_[p] = (unsigned long long*)pf;
+    }
+    HIP_CHECK(hipMemcpy(d_peer_flags_, peer_flags_.data(),
+                        nranks_ * sizeof(void*), hipMemcpyHostToDevice));
+    connected_ = true;
+  }
+
+  int64_t capacity() const { return capacity_; }
+
+  void alltoall(at::Tensor input, at::Tensor output) {
+    py::gil_scoped_release nogil;
+    TORCH_CHECK(connected_, "p2p alltoall used before connect()");
+    check_device_contig(input);
+    check_device_contig(output);
+    int64_t bytes = input.numel() * input.element_size();
+    TORCH_CHECK(output.numel() * output.element_size() == bytes,
+                "p2p alltoall size mismatch");
+    TORCH_CHECK(bytes % nranks_ == 0, "p2p alltoall: not divisible");
+    TORCH_CHECK(bytes <= capacity_, "p2p alltoall: exceeds capacity");
+    int64_t chunk = bytes / nranks_;
+
+    HIP_CHECK(hipMemcpyAsync(send_, input.data_ptr(), bytes,
+                             hipMemcpyDeviceToDevice, stream_));
+    // round barrier: all ranks' send buffers are published
+    ++seq_;
+    bagua_p2p_barrier_launch(d_peer_flags_, (void*)flags_, rank_, nranks_,
+                             seq_, stream_);
+    for (int p = 0; p < nranks_; ++p) {
+      HIP_CHECK(hipMemcpyAsync(
+          (char*)recv_ + p * chunk,
+          (char*)peer_send_[p] + (int64_t)rank_ * chunk, chunk,
+          hipMemcpyDeviceToDevice, stream_));
+    }
+    // completion barrier: nobody may overwrite their send buffer until
+    // every peer finished pulling this round
+    ++seq_;
+    bagua_p2p_barrier_launch(d_peer_flags_, (void*)flags_, rank_, nranks_,
+                             seq_, stream_);
+    HIP_CHECK(hipMemcpyAsync(output.data_ptr(), recv_, bytes,
+                             hipMemcpyDeviceToDevice, stream_));
+  }
+
+ private:
+  int rank_, nranks_;
+  hipStream_t stream_;
+  int64_t capacity_;
+  void* send_ = nullptr;
+  void* recv_ = nullptr;
+  unsigned long long* flags_ = nullptr;
+  void* d_peer_flags_ = nullptr;
+  std::vector<void*> peer_send_;
+  std::vector<unsigned long long*> peer_flags_;
+  unsigned long long seq_ = 0;
+  bool connected_ = false;
+};
+
+// ---------------------------------------------------------------------------
 // kernel wrappers (run on the CALLER's current torch stream so they are
 // ordered with collectives when invoked under torch.cuda.stream(comm))
 // ---------------------------------------------------------------------------
@@ -1018,6 +1156,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("wait_pending", &BucketExecutor::wait_pending)
       .def("synchronize", &BucketExecutor::synchronize)
       .def("clear_buckets", &BucketExecutor::clear_buckets);
+
+  py::class_<P2PAlltoAll>(m, "P2PAlltoAll")
+      .def(py::init<int, int, uintptr_t, int64_t>(), py::arg("rank"),
+           py::arg("nranks"), py::arg("stream"), py::arg("capacity"))
+      .def("handles", &P2PAlltoAll::handles)
+      .def("connect", &P2PAlltoAll::connect)
+      .def("capacity", &P2PAlltoAll::capacity)
+      .def("alltoall", &P2PAlltoAll::alltoall);
 
   m.def("nccl_unique_id", &nccl_unique_id);
   m.def("average_inplace", &average_inplace);

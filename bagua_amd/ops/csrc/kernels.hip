@@ -717,3 +717,47 @@ void bagua_fused_adam_launch(float* p, const float* g, float* m, float* v,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// p2p alltoall support: cross-GPU flag barrier over xGMI.
+//
+// The 8-GPU MI355X node is FULLY CONNECTED point-to-point (7 links/GPU),
+// so a one-hop alltoall is: every rank pulls its chunk from each peer's
+// exported send buffer. Ordering needs a device-side barrier — IPC
+// events re-recorded every round race (a peer's wait can capture the
+// previous recording), so ranks rendezvous on a monotonically increasing
+// sequence number in FINE-GRAINED peer-visible memory instead:
+//   arrive: write seq into MY slot of every peer's flag array
+//   wait:   spin until every slot of MY flag array reaches seq
+// One wave handles all peers (nranks <= 64 by construction).
+// ---------------------------------------------------------------------------
+
+__global__ void p2p_barrier_kernel(
+    unsigned long long* const* __restrict__ peer_flags,  // [nranks] ptrs
+    volatile unsigned long long* __restrict__ my_flags,  // [nranks]
+    int rank, int nranks, unsigned long long seq) {
+  const int p = threadIdx.x;
+  if (p >= nranks) return;
+  // make every prior write to my send buffer visible system-wide before
+  // announcing arrival
+  __threadfence_system();
+  volatile unsigned long long* slot =
+      (volatile unsigned long long*)(peer_flags[p]) + rank;
+  *slot = seq;
+  __threadfence_system();
+  while (my_flags[p] < seq) {
+    __builtin_amdgcn_s_sleep(8);
+  }
+  __threadfence_system();
+}
+
+extern "C" void bagua_p2p_barrier_launch(void* peer_flags_dev_array,
+                                         void* my_flags, int rank,
+                                         int nranks,
+                                         unsigned long long seq,
+                                         hipStream_t stream) {
+  hipLaunchKernelGGL(p2p_barrier_kernel, dim3(1), dim3(64), 0, stream,
+                     (unsigned long long* const*)peer_flags_dev_array,
+                     (volatile unsigned long long*)my_flags, rank, nranks,
+                     seq);
+}

@@ -9,10 +9,13 @@ prints a JSON summary at the end.
 
 import json
 import os
+import sys
 import time
 
 import torch
 import torch.nn.functional as F
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def main():

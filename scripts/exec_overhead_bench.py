@@ -24,6 +24,8 @@ import time
 import torch
 import torch.nn.functional as F
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 
 def run(algo, native, steps=25, warmup=8):
     os.environ["BAGUA_NATIVE_SCHEDULER"] = "1" if native else "0"

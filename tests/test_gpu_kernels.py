@@ -199,3 +199,43 @@ def test_c_abi_roundtrip_single_rank():
     assert torch.equal(x, ref)
     assert torch.equal(y, ref), "alltoall self-copy mismatch"
     lib.bagua_comm_destroy(c)
+
+
+@requires_gpu
+def test_p2p_alltoall_world1():
+    """P2PAlltoAll degenerate path at world 1: allocation, flag barrier
+    kernel, self pull. The 8-GPU one-hop exchange reuses exactly this
+    machinery with IPC-opened peer pointers (ops/p2p.py)."""
+    import os
+
+    os.environ.setdefault("WORLD_SIZE", "1")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("LOCAL_RANK", "0")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29582")
+
+    import bagua_amd
+    from bagua_amd.ops import p2p
+
+    torch.cuda.set_device(0)
+    bagua_amd.init_process_group()
+    comm = bagua_amd.communication._get_default_group() \
+        .get_global_communicator()
+    comm.ensure_native()
+
+    impl = p2p.get_for_communicator(comm, 1 << 20)
+    x = torch.randn(1 << 18, device="cuda")
+    ref = x.clone()
+    out = torch.empty_like(x)
+    with torch.cuda.stream(comm.stream):
+        impl.alltoall(x, out)
+        impl.alltoall(out, out)  # second round exercises seq progression
+    comm.stream.synchronize()
+    assert torch.equal(out, ref)
+    # growth path: larger message forces a new epoch/instance
+    big = torch.randn(1 << 21, device="cuda")
+    impl2 = p2p.get_for_communicator(comm, big.numel() * 4)
+    with torch.cuda.stream(comm.stream):
+        impl2.alltoall(big, big)
+    comm.stream.synchronize()
+    assert torch.isfinite(big).all()
