@@ -186,7 +186,34 @@ class BaguaBackend:
             self.exec_host_ns += time.perf_counter_ns() - t0
             self.exec_dispatches += 1
 
+    def _is_identity_at_world1(self, bucket: BaguaBucket) -> bool:
+        """True iff every op on the bucket is a mathematical no-op at its
+        group's world size 1 (uncompressed centralized sync: allreduce /
+        alltoall+reduce+allgather over one rank leave the bucket bitwise
+        unchanged; compression is NOT identity — it applies quantization
+        noise exactly as at N>1)."""
+        from .bucket import CentralizedSyncOp
+
+        for op in bucket.ops:
+            if not isinstance(op, CentralizedSyncOp):
+                return False
+            if op.compression is not None:
+                return False
+            group = op.group or self.group
+            if len(group.ranks) != 1:
+                return False
+        return True
+
     def _execute_inner(self, bucket: BaguaBucket):
+        if bucket.ops and self._is_identity_at_world1(bucket):
+            # single-rank fast path: skip the event fencing ceremony too
+            # (measured ~0.4 ms/step of host work on VGG16's 9 buckets);
+            # just recycle the ready events
+            for t in bucket.tensors:
+                if t.ready_event is not None:
+                    self.event_pool.put(t.ready_event)
+                    t.ready_event = None
+            return
         if getattr(bucket, "_native_idx", None) is not None:
             events = []
             for t in bucket.tensors:
