@@ -26,8 +26,8 @@ def main():
     os.environ.setdefault("MASTER_PORT", "29599")
     os.environ["BAGUA_AUTOTUNE"] = "1"
     os.environ["BAGUA_AUTOTUNE_WARMUP_TIME_S"] = "3"
-    os.environ["BAGUA_AUTOTUNE_SAMPLING_CONFIDENCE_TIME_S"] = "0.5"
-    os.environ["BAGUA_AUTOTUNE_MAX_SAMPLES"] = "10"
+    os.environ["BAGUA_AUTOTUNE_SAMPLING_CONFIDENCE_TIME_S"] = "0.2"
+    os.environ["BAGUA_AUTOTUNE_MAX_SAMPLES"] = "8"
     os.environ["BAGUA_AUTOTUNE_INTERVAL"] = "25"
 
     torch.cuda.set_device(0)
@@ -64,7 +64,7 @@ def main():
     torch.cuda.synchronize()
 
     windows = []
-    total_steps = 450
+    total_steps = 700
     win = 25
     for w in range(total_steps // win):
         torch.cuda.synchronize()
