@@ -70,6 +70,12 @@ class BaguaBackend:
         # many bucket dispatches it covers
         self.exec_host_ns = 0
         self.exec_dispatches = 0
+        # optional per-op trace (BAGUA_TRACE=1): ring buffer of
+        # (bucket, op kinds, host start/end ns) — the per-op tracing
+        # depth of the reference's tracing crate, dumpable as a
+        # chrome://tracing file via dump_trace()
+        self.trace_enabled = os.environ.get("BAGUA_TRACE", "0") == "1"
+        self.trace_events: List[dict] = []
 
     # ------------------------------------------------------------------
     def register_ordered_buckets(self, buckets: List[BaguaBucket]):
@@ -183,8 +189,24 @@ class BaguaBackend:
         finally:
             if _ROCTX:
                 torch.cuda.nvtx.range_pop()
-            self.exec_host_ns += time.perf_counter_ns() - t0
+            end = time.perf_counter_ns()
+            self.exec_host_ns += end - t0
             self.exec_dispatches += 1
+            if self.trace_enabled and len(self.trace_events) < 100000:
+                self.trace_events.append({
+                    "name": "bucket:%s" % bucket.name,
+                    "cat": ",".join(type(op).__name__ for op in bucket.ops),
+                    "ph": "X", "pid": env.get_rank(), "tid": 0,
+                    "ts": t0 / 1e3, "dur": (end - t0) / 1e3})
+
+    def dump_trace(self, path: str):
+        """Write recorded spans as a chrome://tracing / Perfetto JSON
+        file (enable with BAGUA_TRACE=1)."""
+        import json
+
+        with open(path, "w") as f:
+            json.dump({"traceEvents": self.trace_events,
+                       "displayTimeUnit": "ms"}, f)
 
     def _is_identity_at_world1(self, bucket: BaguaBucket) -> bool:
         """True iff every op on the bucket is a mathematical no-op at its
