@@ -103,6 +103,24 @@ def compress_chunked(flat: torch.Tensor, num_chunks: int,
     return quant.compress_chunked(flat, num_chunks, target_chunk, out)
 
 
+def dequant_reduce(buf: torch.Tensor, flat: torch.Tensor, num_chunks: int,
+                   target_chunk: int, average: bool):
+    """Fused: flat[target_chunk] = reduce over the dequantized wire
+    chunks — skips materializing the non-target chunks (they are only
+    reduction inputs on the ByteGrad path). Bitwise-identical to
+    decompress_chunked_into + reduce_chunk_inplace: values round through
+    flat.dtype between dequantize and f32 accumulation."""
+    if _use_native(buf, flat):
+        native.lib().dequant_reduce(buf, flat, num_chunks, target_chunk,
+                                    average)
+        return
+    chunk = flat.numel() // num_chunks
+    dec = quant.decompress_chunked(buf, num_chunks, chunk, flat.dtype)
+    v = dec.view(num_chunks, -1).float()
+    red = v.mean(0) if average else v.sum(0)
+    flat.view(num_chunks, -1)[target_chunk].copy_(red.to(flat.dtype))
+
+
 def decompress_chunked_into(buf: torch.Tensor, flat: torch.Tensor,
                             num_chunks: int, target_chunk: int = -1):
     """Decompress the wire buffer into ``flat`` (all chunks or one)."""

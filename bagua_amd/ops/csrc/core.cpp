@@ -46,6 +46,10 @@ void bagua_decompress_launch(int dtype, const uint8_t* in, void* x,
                              size_t chunk, size_t chunk_stride,
                              int chunk_begin, int chunk_count,
                              hipStream_t stream);
+void bagua_dequant_reduce_launch(int dtype, const uint8_t* in, void* x,
+                                 size_t chunk, size_t chunk_stride,
+                                 int num_chunks, int target_chunk,
+                                 int average, hipStream_t stream);
 void bagua_fused_sgd_launch(float* p, const float* g, float* m, float lr,
                             float momentum, float dampening,
                             float weight_decay, int nesterov,
@@ -567,10 +571,14 @@ class BucketExecutor {
     comm->alltoall(b.wire, b.wire_tmp);
     std::swap(b.wire, b.wire_tmp);
     wire = (uint8_t*)b.wire.data_ptr();
-    bagua_decompress_launch(dt, wire, b.flat.data_ptr(), chunk, stride, 0,
-                            n, stream_);
-    bagua_reduce_chunk_launch(dt, b.flat.data_ptr(), n, rank,
-                              b.average ? 1 : 0, chunk, stream_);
+    // fused dequantize+reduce straight from the wire into the target
+    // chunk: the non-target chunks were only reduction inputs, so the
+    // old decompress-then-reduce pair paid ~2 extra bucket passes of
+    // HBM traffic for nothing (bitwise-identical result — the kernel
+    // rounds through T between dequantize and accumulate)
+    bagua_dequant_reduce_launch(dt, wire, b.flat.data_ptr(), chunk,
+                                stride, n, rank, b.average ? 1 : 0,
+                                stream_);
     bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, partials,
                           chunk, stride, n, rank, 1, stream_);
     // in-place allgather of the rank's wire chunk (identity at n==1)
@@ -858,6 +866,23 @@ static void decompress_chunked(at::Tensor buf, at::Tensor flat,
   bagua_decompress_launch(kernel_dtype(flat), (uint8_t*)buf.data_ptr(),
                           flat.data_ptr(), chunk, stride, begin, count,
                           current_stream());
+}
+
+// fused: flat[target_chunk] = reduce over dequantized wire chunks
+static void dequant_reduce(at::Tensor buf, at::Tensor flat,
+                           int64_t num_chunks, int64_t target_chunk,
+                           bool average) {
+  check_device_contig(flat);
+  check_device_contig(buf);
+  TORCH_CHECK(flat.numel() % num_chunks == 0, "chunked size mismatch");
+  TORCH_CHECK(num_chunks <= 64, "dequant_reduce supports <= 64 chunks");
+  int64_t chunk = flat.numel() / num_chunks;
+  int64_t stride = compressed_chunk_stride(chunk);
+  bagua_dequant_reduce_launch(kernel_dtype(flat),
+                              (uint8_t*)buf.data_ptr(), flat.data_ptr(),
+                              chunk, stride, (int)num_chunks,
+                              (int)target_chunk, average ? 1 : 0,
+                              current_stream());
 }
 
 static void fused_sgd_step(at::Tensor p, at::Tensor g, at::Tensor m,
@@ -1175,6 +1200,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_chunk_inplace", &reduce_chunk_inplace);
   m.def("compress_chunked", &compress_chunked);
   m.def("decompress_chunked", &decompress_chunked);
+  m.def("dequant_reduce", &dequant_reduce);
   m.def("compressed_chunk_stride", &compressed_chunk_stride);
   m.def("fused_sgd_step", &fused_sgd_step);
   m.def("fused_sgd_mixed_step", &fused_sgd_mixed_step);

@@ -385,6 +385,80 @@ __global__ void dequantize_kernel(const uint8_t* __restrict__ in,
 }
 
 // ---------------------------------------------------------------------------
+// fused dequantize + chunk reduce (ByteGrad hot path).
+//
+// After the alltoall, the bucket's non-target chunks exist only as
+// inputs to the reduction — materializing them through a full
+// decompress pass and re-reading them in reduce_chunk wastes ~2 bucket
+// passes of HBM traffic. This kernel reads the wire payloads directly
+// (n_chunks x u8), dequantizes in registers and writes ONLY the target
+// chunk. Values round through T between dequantize and accumulate so
+// the result is BITWISE identical to dequantize_kernel +
+// reduce_chunk_kernel (the CPU oracle and the python executor take that
+// unfused chain).
+// ---------------------------------------------------------------------------
+
+#define DQR_MAX_CHUNKS 64
+
+template <typename T>
+__global__ void dequant_reduce_kernel(const uint8_t* __restrict__ in,
+                                      size_t chunk, size_t chunk_stride,
+                                      int num_chunks, float post_scale,
+                                      T* __restrict__ dst /* target base */) {
+  // per-chunk quant params staged through LDS once per block
+  __shared__ float s_lower[DQR_MAX_CHUNKS];
+  __shared__ float s_inv_scale[DQR_MAX_CHUNKS];
+  for (int c = threadIdx.x; c < num_chunks; c += blockDim.x) {
+    const float* hdr =
+        reinterpret_cast<const float*>(in + (size_t)c * chunk_stride);
+    float scale, lower, upper;
+    quant_params(hdr[0], hdr[1], scale, lower, upper);
+    s_lower[c] = lower;
+    s_inv_scale[c] = 1.0f / scale;
+  }
+  __syncthreads();
+
+  constexpr int V = Vec16<T>::N;
+  using VT = Vec16<T>;
+  const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const size_t nv = chunk / V;
+  VT* dv = reinterpret_cast<VT*>(dst);
+  for (size_t i = tid; i < nv; i += stride) {
+    float acc[Vec16<T>::N];
+#pragma unroll
+    for (int k = 0; k < V; ++k) acc[k] = 0.f;
+    for (int c = 0; c < num_chunks; ++c) {
+      const uint8_t* payload = in + (size_t)c * chunk_stride + 32;
+      uint8_t q[Vec16<T>::N];
+      if (V == 4)
+        *reinterpret_cast<uint32_t*>(q) =
+            *reinterpret_cast<const uint32_t*>(payload + i * 4);
+      else
+        *reinterpret_cast<uint64_t*>(q) =
+            *reinterpret_cast<const uint64_t*>(payload + i * 8);
+      const float lower = s_lower[c], inv_scale = s_inv_scale[c];
+#pragma unroll
+      for (int k = 0; k < V; ++k)
+        acc[k] += to_f(from_f<T>(((float)q[k] + lower) * inv_scale));
+    }
+    VT out;
+#pragma unroll
+    for (int k = 0; k < V; ++k) out.v[k] = from_f<T>(acc[k] * post_scale);
+    dv[i] = out;
+  }
+  for (size_t i = nv * V + tid; i < chunk; i += stride) {
+    float acc = 0.f;
+    for (int c = 0; c < num_chunks; ++c) {
+      const uint8_t* payload = in + (size_t)c * chunk_stride + 32;
+      acc += to_f(from_f<T>(((float)payload[i] + s_lower[c])
+                            * s_inv_scale[c]));
+    }
+    dst[i] = from_f<T>(acc * post_scale);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // host-side launchers (extern "C", stream-passing; no syncs)
 // ---------------------------------------------------------------------------
 
@@ -558,6 +632,36 @@ void bagua_decompress_launch(int dtype, const uint8_t* in, void* x,
       hipLaunchKernelGGL((dequantize_kernel<__hip_bfloat16>), grid,
                          dim3(BLOCK), 0, stream, in, chunk, chunk_begin,
                          chunk_stride, (__hip_bfloat16*)x);
+      break;
+  }
+}
+
+// fused dequantize + reduce of all wire chunks into flat's target chunk
+// (x points at the BUCKET base; target offset applied here)
+void bagua_dequant_reduce_launch(int dtype, const uint8_t* in, void* x,
+                                 size_t chunk, size_t chunk_stride,
+                                 int num_chunks, int target_chunk,
+                                 int average, hipStream_t stream) {
+  float post_scale = average ? 1.0f / (float)num_chunks : 1.0f;
+  int grid = grid_for(chunk / 8 + 1);
+  switch (dtype) {
+    case 0:
+      hipLaunchKernelGGL((dequant_reduce_kernel<float>), dim3(grid),
+                         dim3(BLOCK), 0, stream, in, chunk, chunk_stride,
+                         num_chunks, post_scale,
+                         (float*)x + (size_t)target_chunk * chunk);
+      break;
+    case 1:
+      hipLaunchKernelGGL((dequant_reduce_kernel<__half>), dim3(grid),
+                         dim3(BLOCK), 0, stream, in, chunk, chunk_stride,
+                         num_chunks, post_scale,
+                         (__half*)x + (size_t)target_chunk * chunk);
+      break;
+    case 2:
+      hipLaunchKernelGGL((dequant_reduce_kernel<__hip_bfloat16>),
+                         dim3(grid), dim3(BLOCK), 0, stream, in, chunk,
+                         chunk_stride, num_chunks, post_scale,
+                         (__hip_bfloat16*)x + (size_t)target_chunk * chunk);
       break;
   }
 }

@@ -239,3 +239,33 @@ def test_p2p_alltoall_world1():
         impl2.alltoall(big, big)
     comm.stream.synchronize()
     assert torch.isfinite(big).all()
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float16,
+                                   torch.bfloat16])
+@pytest.mark.parametrize("num_chunks", [2, 8])
+def test_dequant_reduce_matches_unfused(dtype, num_chunks):
+    """Fused dequantize+reduce must be BITWISE identical to the
+    decompress -> reduce_chunk chain (it rounds through T between
+    dequantize and f32 accumulation by construction)."""
+    from bagua_amd import ops
+
+    torch.manual_seed(11)
+    chunk = 4096
+    flat = (torch.randn(num_chunks * chunk, device="cuda") * 3).to(dtype)
+    wire = ops.compress_chunked(flat, num_chunks)
+
+    a = flat.clone()
+    ops.decompress_chunked_into(wire, a, num_chunks)
+    ops.reduce_chunk_inplace(a, num_chunks, 1, True)
+
+    b = flat.clone()
+    ops.dequant_reduce(wire, b, num_chunks, 1, True)
+    torch.cuda.synchronize()
+
+    av = a.view(num_chunks, -1)[1]
+    bv = b.view(num_chunks, -1)[1]
+    assert torch.equal(av, bv), (
+        "fused dequant_reduce deviates (max diff %g)"
+        % (av.float() - bv.float()).abs().max().item())
