@@ -66,6 +66,21 @@ def main():
             n + n * es)
         results.append(("decompress_chunked", name, gbs, us))
 
+        # fused dequantize+reduce (ByteGrad hot path) vs the unfused
+        # chain it replaces: decompress (n + n*es) + reduce
+        # (n*es + chunk*es) -> fused reads n u8 + writes chunk*es only
+        gbs, us = bench(
+            lambda: lib.dequant_reduce(wire, x, chunks, 0, True),
+            n + chunk * es)
+        results.append(("dequant_reduce(8)", name, gbs, us))
+
+        def unfused():
+            lib.decompress_chunked(wire, x, chunks, -1)
+            lib.reduce_chunk_inplace(x, chunks, 0, True)
+
+        gbs, us = bench(unfused, n + chunk * es)  # same useful bytes
+        results.append(("  unfused chain", name, gbs, us))
+
     m = torch.zeros(n, device="cuda")
     g = torch.randn(n, device="cuda")
     p = torch.randn(n, device="cuda")
