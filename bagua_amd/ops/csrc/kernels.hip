@@ -405,6 +405,11 @@ __global__ void dequant_reduce_kernel(const uint8_t* __restrict__ in,
                                       size_t chunk, size_t chunk_stride,
                                       int num_chunks, float post_scale,
                                       T* __restrict__ dst /* target base */) {
+  // no FMA contraction: for T=float the dequantized value must round
+  // exactly as the unfused dequantize kernel's store did, or bitwise
+  // parity with decompress+reduce breaks (memory-bound kernel — FMA
+  // contributes nothing here anyway)
+#pragma clang fp contract(off)
   // per-chunk quant params staged through LDS once per block
   __shared__ float s_lower[DQR_MAX_CHUNKS];
   __shared__ float s_inv_scale[DQR_MAX_CHUNKS];
