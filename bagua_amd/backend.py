@@ -120,6 +120,8 @@ class BaguaBackend:
         if os.environ.get("BAGUA_NATIVE_SCHEDULER", "1") != "1":
             return
 
+        from .ops import p2p
+
         def eligible(b: BaguaBucket) -> bool:
             if b._flat is None or len(b.ops) != 1:
                 return False
@@ -127,6 +129,12 @@ class BaguaBackend:
             if not isinstance(op, CentralizedSyncOp):
                 return False
             if op.compression not in (None, "MinMaxUInt8"):
+                return False
+            # the opt-in p2p alltoall lives on the python communicator;
+            # route alltoall-shaped ops there so the flag means what it
+            # says instead of silently using RCCL in the C++ executor
+            if p2p.enabled() and (op.compression is not None
+                                  or op.scattergather):
                 return False
             return b._flat.dtype in (torch.float32, torch.float16,
                                      torch.bfloat16)
