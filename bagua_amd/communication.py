@@ -258,7 +258,26 @@ class BaguaCommunicator:
 
         self._run(native, fallback, [send_tensor, recv_tensor])
 
+    def _p2p_allgather_inplace(self, tensor) -> bool:
+        """Opt-in one-hop xGMI allgather (same machinery as the p2p
+        alltoall: each rank pulls peer p's own chunk directly)."""
+        from .ops import p2p
+
+        if (not p2p.enabled() or self.nranks() <= 1
+                or not _is_cuda_job() or not tensor.is_cuda):
+            return False
+        impl = p2p.get_for_communicator(
+            self, tensor.numel() * tensor.element_size())
+
+        def native():
+            impl.allgather_inplace(tensor)
+
+        self._run(native, lambda: None, [tensor])
+        return True
+
     def allgather_inplace(self, tensor):
+        if self._p2p_allgather_inplace(tensor):
+            return
         """tensor is the full buffer; rank's own chunk is the input."""
         n = self.nranks()
         chunk = tensor.numel() // n

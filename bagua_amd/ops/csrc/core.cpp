@@ -737,26 +737,50 @@ class P2PAlltoAll {
 
   void alltoall(at::Tensor input, at::Tensor output) {
     py::gil_scoped_release nogil;
-    TORCH_CHECK(connected_, "p2p alltoall used before connect()");
+    exchange_(input, output, /*allgather=*/false);
+  }
+
+  // one-hop allgather: every rank publishes ONLY its own chunk (at its
+  // rank offset in the exported buffer) and pulls chunk p from peer p's
+  // own-chunk slot. input/output are the FULL buffer; input's rank-chunk
+  // is the contribution (ncclAllGather-inplace shape).
+  void allgather_inplace(at::Tensor t) {
+    py::gil_scoped_release nogil;
+    exchange_(t, t, /*allgather=*/true);
+  }
+
+ private:
+  void exchange_(at::Tensor input, at::Tensor output, bool allgather) {
+    TORCH_CHECK(connected_, "p2p exchange used before connect()");
     check_device_contig(input);
     check_device_contig(output);
     int64_t bytes = input.numel() * input.element_size();
     TORCH_CHECK(output.numel() * output.element_size() == bytes,
-                "p2p alltoall size mismatch");
-    TORCH_CHECK(bytes % nranks_ == 0, "p2p alltoall: not divisible");
-    TORCH_CHECK(bytes <= capacity_, "p2p alltoall: exceeds capacity");
+                "p2p exchange size mismatch");
+    TORCH_CHECK(bytes % nranks_ == 0, "p2p exchange: not divisible");
+    TORCH_CHECK(bytes <= capacity_, "p2p exchange: exceeds capacity");
     int64_t chunk = bytes / nranks_;
 
-    HIP_CHECK(hipMemcpyAsync(send_, input.data_ptr(), bytes,
-                             hipMemcpyDeviceToDevice, stream_));
+    if (allgather) {
+      // publish my own chunk only
+      HIP_CHECK(hipMemcpyAsync(
+          (char*)send_ + (int64_t)rank_ * chunk,
+          (char*)input.data_ptr() + (int64_t)rank_ * chunk, chunk,
+          hipMemcpyDeviceToDevice, stream_));
+    } else {
+      HIP_CHECK(hipMemcpyAsync(send_, input.data_ptr(), bytes,
+                               hipMemcpyDeviceToDevice, stream_));
+    }
     // round barrier: all ranks' send buffers are published
     ++seq_;
     bagua_p2p_barrier_launch(d_peer_flags_, (void*)flags_, rank_, nranks_,
                              seq_, stream_);
     for (int p = 0; p < nranks_; ++p) {
+      // alltoall pulls MY slice of peer p; allgather pulls peer p's OWN
+      // chunk
+      int64_t src_off = (allgather ? (int64_t)p : (int64_t)rank_) * chunk;
       HIP_CHECK(hipMemcpyAsync(
-          (char*)recv_ + p * chunk,
-          (char*)peer_send_[p] + (int64_t)rank_ * chunk, chunk,
+          (char*)recv_ + p * chunk, (char*)peer_send_[p] + src_off, chunk,
           hipMemcpyDeviceToDevice, stream_));
     }
     // completion barrier: nobody may overwrite their send buffer until
@@ -767,6 +791,8 @@ class P2PAlltoAll {
     HIP_CHECK(hipMemcpyAsync(output.data_ptr(), recv_, bytes,
                              hipMemcpyDeviceToDevice, stream_));
   }
+
+ public:
 
  private:
   int rank_, nranks_;
@@ -1188,7 +1214,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("handles", &P2PAlltoAll::handles)
       .def("connect", &P2PAlltoAll::connect)
       .def("capacity", &P2PAlltoAll::capacity)
-      .def("alltoall", &P2PAlltoAll::alltoall);
+      .def("alltoall", &P2PAlltoAll::alltoall)
+      .def("allgather_inplace", &P2PAlltoAll::allgather_inplace);
 
   m.def("nccl_unique_id", &nccl_unique_id);
   m.def("average_inplace", &average_inplace);

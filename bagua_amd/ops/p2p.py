@@ -56,6 +56,9 @@ class P2pAllToAll:
     def alltoall(self, t_in, t_out):
         self.impl.alltoall(t_in, t_out)
 
+    def allgather_inplace(self, t):
+        self.impl.allgather_inplace(t)
+
 
 def get_for_communicator(comm, bytes_needed: int):
     """Cached, lazily grown instance on a BaguaCommunicator. Collective:

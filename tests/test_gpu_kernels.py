@@ -239,6 +239,14 @@ def test_p2p_alltoall_world1():
         impl2.alltoall(big, big)
     comm.stream.synchronize()
     assert torch.isfinite(big).all()
+    # allgather leg (one-hop pull of each peer's own chunk; identity at
+    # world 1)
+    g = torch.randn(1 << 18, device="cuda")
+    gref = g.clone()
+    with torch.cuda.stream(comm.stream):
+        impl2.allgather_inplace(g)
+    comm.stream.synchronize()
+    assert torch.equal(g, gref)
 
 
 @requires_gpu
