@@ -208,3 +208,52 @@ def test_world8_bench_contract():
     rec = json.loads(json_lines[0])
     assert rec["n_gpus"] == 8
     assert rec["value"] > 0
+
+
+def _worker_moe8(rank, nprocs):
+    import torch.nn as nn
+
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+    from bagua_amd.parallel.moe import is_moe_param
+    from tests.test_moe import MoEModel
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)
+    model = MoEModel(k=1, num_local_experts=1)  # 8 experts at world 8
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm())
+    assert model.moe.num_experts == nprocs
+    for step in range(4):
+        torch.manual_seed(700 + rank * 13 + step)
+        data = torch.randn(16, 8)
+        target = torch.randn(16, 4)
+        optimizer.zero_grad()
+        out, l_aux = ddp(data)
+        loss = F.mse_loss(out, target) + 0.01 * l_aux
+        loss.backward()
+        optimizer.step()
+        assert torch.isfinite(loss)
+    dense = torch.cat([p.detach().reshape(-1)
+                       for n, p in sorted(model.named_parameters())
+                       if not is_moe_param(p)])
+    expert = torch.cat([p.detach().reshape(-1)
+                        for n, p in sorted(model.named_parameters())
+                        if is_moe_param(p)])
+    bagua_amd.deinit_process_group()
+    return dense, expert
+
+
+def test_world8_moe_expert_parallel():
+    """EP over the full 8-rank alltoall: dense consensus, expert shards
+    stay rank-local (the driver's world size)."""
+    out = run_multi_process(WORLD, _worker_moe8, timeout=420)
+    for r in range(1, WORLD):
+        assert torch.equal(out[0][0], out[r][0]), "dense diverged"
+    for r in range(1, WORLD):
+        assert not torch.equal(out[0][1], out[r][1]), (
+            "experts identical across EP ranks")
