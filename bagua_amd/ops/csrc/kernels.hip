@@ -423,34 +423,40 @@ __global__ void dequant_reduce_kernel(const uint8_t* __restrict__ in,
   }
   __syncthreads();
 
-  constexpr int V = Vec16<T>::N;
+  // 16 elements per lane: ONE 16-byte u8 payload load per chunk per
+  // iteration (the CDNA4 coalescing sweet spot — the per-T Vec16 width
+  // would shrink payload loads to 4/8 bytes), then 16*sizeof(T)/16
+  // 16-byte stores out.
+  constexpr int V = 16;
+  constexpr int OV = Vec16<T>::N;       // elems per 16B output vector
+  constexpr int NOUT = V / OV;          // output vectors per iteration
   using VT = Vec16<T>;
   const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   const size_t nv = chunk / V;
   VT* dv = reinterpret_cast<VT*>(dst);
   for (size_t i = tid; i < nv; i += stride) {
-    float acc[Vec16<T>::N];
+    float acc[V];
 #pragma unroll
     for (int k = 0; k < V; ++k) acc[k] = 0.f;
     for (int c = 0; c < num_chunks; ++c) {
       const uint8_t* payload = in + (size_t)c * chunk_stride + 32;
-      uint8_t q[Vec16<T>::N];
-      if (V == 4)
-        *reinterpret_cast<uint32_t*>(q) =
-            *reinterpret_cast<const uint32_t*>(payload + i * 4);
-      else
-        *reinterpret_cast<uint64_t*>(q) =
-            *reinterpret_cast<const uint64_t*>(payload + i * 8);
+      alignas(16) uint8_t q[V];
+      *reinterpret_cast<uint4*>(q) =
+          *reinterpret_cast<const uint4*>(payload + i * V);
       const float lower = s_lower[c], inv_scale = s_inv_scale[c];
 #pragma unroll
       for (int k = 0; k < V; ++k)
         acc[k] += to_f(from_f<T>(((float)q[k] + lower) * inv_scale));
     }
-    VT out;
 #pragma unroll
-    for (int k = 0; k < V; ++k) out.v[k] = from_f<T>(acc[k] * post_scale);
-    dv[i] = out;
+    for (int o = 0; o < NOUT; ++o) {
+      VT out;
+#pragma unroll
+      for (int k = 0; k < OV; ++k)
+        out.v[k] = from_f<T>(acc[o * OV + k] * post_scale);
+      dv[i * NOUT + o] = out;
+    }
   }
   for (size_t i = nv * V + tid; i < chunk; i += stride) {
     float acc = 0.f;
@@ -648,7 +654,7 @@ void bagua_dequant_reduce_launch(int dtype, const uint8_t* in, void* x,
                                  int num_chunks, int target_chunk,
                                  int average, hipStream_t stream) {
   float post_scale = average ? 1.0f / (float)num_chunks : 1.0f;
-  int grid = grid_for(chunk / 8 + 1);
+  int grid = grid_for(chunk / 16 + 1);
   switch (dtype) {
     case 0:
       hipLaunchKernelGGL((dequant_reduce_kernel<float>), dim3(grid),
