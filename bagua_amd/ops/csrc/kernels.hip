@@ -317,27 +317,29 @@ __global__ void quantize_kernel(const T* __restrict__ x, size_t chunk,
   const T* __restrict__ src = x + (size_t)c * chunk;
   uint8_t* __restrict__ payload = dst + 32;
 
-  constexpr int V = Vec16<T>::N;  // elements per 16B load; V u8 out
+  // 16 elements per lane: 16*sizeof(T)/16 16-byte input loads, ONE
+  // 16-byte u8 store (the per-T width stored only 4/8 bytes per lane)
+  constexpr int V = 16;
+  constexpr int IV = Vec16<T>::N;
+  constexpr int NIN = V / IV;
   using VT = Vec16<T>;
   const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   const size_t nv = chunk / V;
   const VT* sv = reinterpret_cast<const VT*>(src);
   for (size_t i = tid; i < nv; i += stride) {
-    VT a = sv[i];
-    uint8_t q[Vec16<T>::N];
+    alignas(16) uint8_t q[V];
 #pragma unroll
-    for (int k = 0; k < V; ++k) {
-      float level = fminf(rintf(to_f(a.v[k]) * scale), upper);
-      q[k] = (uint8_t)(level - lower);
+    for (int in_v = 0; in_v < NIN; ++in_v) {
+      VT a = sv[i * NIN + in_v];
+#pragma unroll
+      for (int k = 0; k < IV; ++k) {
+        float level = fminf(rintf(to_f(a.v[k]) * scale), upper);
+        q[in_v * IV + k] = (uint8_t)(level - lower);
+      }
     }
-    // V is 4 (f32) or 8 (f16/bf16): one 4- or 8-byte store
-    if (V == 4)
-      *reinterpret_cast<uint32_t*>(payload + i * 4) =
-          *reinterpret_cast<uint32_t*>(q);
-    else
-      *reinterpret_cast<uint64_t*>(payload + i * 8) =
-          *reinterpret_cast<uint64_t*>(q);
+    *reinterpret_cast<uint4*>(payload + i * V) =
+        *reinterpret_cast<uint4*>(q);
   }
   for (size_t i = nv * V + tid; i < chunk; i += stride) {
     float level = fminf(rintf(to_f(src[i]) * scale), upper);
@@ -360,25 +362,29 @@ __global__ void dequantize_kernel(const uint8_t* __restrict__ in,
   const uint8_t* __restrict__ payload = src + 32;
   T* __restrict__ dst = x + (size_t)c * chunk;
 
-  constexpr int V = Vec16<T>::N;
+  // 16 elements per lane: one 16-byte payload load, then
+  // 16*sizeof(T)/16 16-byte stores (the per-T Vec16 width would shrink
+  // f32 payload loads to 4 bytes)
+  constexpr int V = 16;
+  constexpr int OV = Vec16<T>::N;
+  constexpr int NOUT = V / OV;
   using VT = Vec16<T>;
   const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   const size_t nv = chunk / V;
   VT* dv = reinterpret_cast<VT*>(dst);
   for (size_t i = tid; i < nv; i += stride) {
-    uint8_t q[Vec16<T>::N];
-    if (V == 4)
-      *reinterpret_cast<uint32_t*>(q) =
-          *reinterpret_cast<const uint32_t*>(payload + i * 4);
-    else
-      *reinterpret_cast<uint64_t*>(q) =
-          *reinterpret_cast<const uint64_t*>(payload + i * 8);
-    VT out;
+    alignas(16) uint8_t q[V];
+    *reinterpret_cast<uint4*>(q) =
+        *reinterpret_cast<const uint4*>(payload + i * V);
 #pragma unroll
-    for (int k = 0; k < V; ++k)
-      out.v[k] = from_f<T>(((float)q[k] + lower) * inv_scale);
-    dv[i] = out;
+    for (int o = 0; o < NOUT; ++o) {
+      VT out;
+#pragma unroll
+      for (int k = 0; k < OV; ++k)
+        out.v[k] = from_f<T>(((float)q[o * OV + k] + lower) * inv_scale);
+      dv[i * NOUT + o] = out;
+    }
   }
   for (size_t i = nv * V + tid; i < chunk; i += stride)
     dst[i] = from_f<T>(((float)payload[i] + lower) * inv_scale);
