@@ -78,26 +78,35 @@ __device__ __forceinline__ float ew_apply(float x, float y, float f) {
 template <typename T, int OP>
 __global__ void ew_kernel(T* __restrict__ x, const T* __restrict__ y,
                           float f, size_t n) {
+  // two 16-byte vectors per lane per iteration (32 B accesses): +9% on
+  // the f32 read-modify-write pattern (round-1 measurement); unlike the
+  // compressor kernels this path has no competing access width.
   constexpr int V = Vec16<T>::N;
   using VT = Vec16<T>;
   const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
-  const size_t nv = n / V;
+  const size_t nv2 = n / (2 * V);
 
   VT* xv = reinterpret_cast<VT*>(x);
   const VT* yv = reinterpret_cast<const VT*>(y);
-  for (size_t i = tid; i < nv; i += stride) {
-    VT a = xv[i];
-    VT b;
-    if (OP != EW_DIV) b = yv[i];
+  for (size_t i = tid; i < nv2; i += stride) {
+    VT a0 = xv[2 * i], a1 = xv[2 * i + 1];
+    VT b0, b1;
+    if (OP != EW_DIV) {
+      b0 = yv[2 * i];
+      b1 = yv[2 * i + 1];
+    }
 #pragma unroll
     for (int k = 0; k < V; ++k) {
-      float bx = (OP != EW_DIV) ? to_f(b.v[k]) : 0.f;
-      a.v[k] = from_f<T>(ew_apply<OP>(to_f(a.v[k]), bx, f));
+      float bx0 = (OP != EW_DIV) ? to_f(b0.v[k]) : 0.f;
+      float bx1 = (OP != EW_DIV) ? to_f(b1.v[k]) : 0.f;
+      a0.v[k] = from_f<T>(ew_apply<OP>(to_f(a0.v[k]), bx0, f));
+      a1.v[k] = from_f<T>(ew_apply<OP>(to_f(a1.v[k]), bx1, f));
     }
-    xv[i] = a;
+    xv[2 * i] = a0;
+    xv[2 * i + 1] = a1;
   }
-  for (size_t i = nv * V + tid; i < n; i += stride) {
+  for (size_t i = nv2 * 2 * V + tid; i < n; i += stride) {
     float bx = (OP != EW_DIV) ? to_f(y[i]) : 0.f;
     x[i] = from_f<T>(ew_apply<OP>(to_f(x[i]), bx, f));
   }
@@ -489,7 +498,7 @@ static inline int grid_for(size_t work_items) {
 template <typename T, int OP>
 static void launch_ew(void* x, const void* y, float f, size_t n,
                       hipStream_t stream) {
-  int grid = grid_for(n / Vec16<T>::N + 1);
+  int grid = grid_for(n / (2 * Vec16<T>::N) + 1);
   hipLaunchKernelGGL((ew_kernel<T, OP>), dim3(grid), dim3(BLOCK), 0, stream,
                      (T*)x, (const T*)y, f, n);
 }
