@@ -52,9 +52,10 @@ template <> __device__ __forceinline__ __hip_bfloat16 from_f<__hip_bfloat16>(flo
   return __float2bfloat16(v);
 }
 
-// 16-byte per-lane vector of T (the coalescing sweet spot; a 32 B f32
-// variant gained +9% on elementwise but cost -26% on f32 decompress,
-// which ByteGrad hits twice per step — 16 B wins overall).
+// 16-byte per-lane vector of T. Access widths are now tuned PER KERNEL:
+// elementwise uses 2x Vec16 (32 B), the compressor kernels use a fixed
+// 16 u8 payload granule (16 B) with 16*sizeof(T)/16 Vec16 tensor-side
+// vectors — the round-1 one-width-fits-all tradeoff is gone.
 template <typename T> struct alignas(16) Vec16 {
   static constexpr int N = 16 / sizeof(T);
   T v[N];
