@@ -327,29 +327,28 @@ __global__ void quantize_kernel(const T* __restrict__ x, size_t chunk,
   const T* __restrict__ src = x + (size_t)c * chunk;
   uint8_t* __restrict__ payload = dst + 32;
 
-  // 16 elements per lane: 16*sizeof(T)/16 16-byte input loads, ONE
-  // 16-byte u8 store (the per-T width stored only 4/8 bytes per lane)
-  constexpr int V = 16;
-  constexpr int IV = Vec16<T>::N;
-  constexpr int NIN = V / IV;
+  // per-T width (measured, gpurun r2c9): 16-byte tensor-side loads win;
+  // widening the u8 store granule beyond that costs ~1%
+  constexpr int V = Vec16<T>::N;  // 4 (f32) / 8 (f16, bf16)
   using VT = Vec16<T>;
   const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   const size_t nv = chunk / V;
   const VT* sv = reinterpret_cast<const VT*>(src);
   for (size_t i = tid; i < nv; i += stride) {
-    alignas(16) uint8_t q[V];
+    VT a = sv[i];
+    uint8_t q[Vec16<T>::N];
 #pragma unroll
-    for (int in_v = 0; in_v < NIN; ++in_v) {
-      VT a = sv[i * NIN + in_v];
-#pragma unroll
-      for (int k = 0; k < IV; ++k) {
-        float level = fminf(rintf(to_f(a.v[k]) * scale), upper);
-        q[in_v * IV + k] = (uint8_t)(level - lower);
-      }
+    for (int k = 0; k < V; ++k) {
+      float level = fminf(rintf(to_f(a.v[k]) * scale), upper);
+      q[k] = (uint8_t)(level - lower);
     }
-    *reinterpret_cast<uint4*>(payload + i * V) =
-        *reinterpret_cast<uint4*>(q);
+    if (V == 4)
+      *reinterpret_cast<uint32_t*>(payload + i * 4) =
+          *reinterpret_cast<uint32_t*>(q);
+    else
+      *reinterpret_cast<uint64_t*>(payload + i * 8) =
+          *reinterpret_cast<uint64_t*>(q);
   }
   for (size_t i = nv * V + tid; i < chunk; i += stride) {
     float level = fminf(rintf(to_f(src[i]) * scale), upper);
@@ -372,10 +371,11 @@ __global__ void dequantize_kernel(const uint8_t* __restrict__ in,
   const uint8_t* __restrict__ payload = src + 32;
   T* __restrict__ dst = x + (size_t)c * chunk;
 
-  // 16 elements per lane: one 16-byte payload load, then
-  // 16*sizeof(T)/16 16-byte stores (the per-T Vec16 width would shrink
-  // f32 payload loads to 4 bytes)
-  constexpr int V = 16;
+  // per-dtype width (measured, gpurun r2c9): f32 wants V=4 (4-byte
+  // payload load + one 16-byte store — V=16's four stores per lane cost
+  // -43%); f16/bf16 want V=16 (one 16-byte payload load, two 16-byte
+  // stores — +21% over V=8)
+  constexpr int V = sizeof(T) == 4 ? 4 : 16;
   constexpr int OV = Vec16<T>::N;
   constexpr int NOUT = V / OV;
   using VT = Vec16<T>;
@@ -385,8 +385,12 @@ __global__ void dequantize_kernel(const uint8_t* __restrict__ in,
   VT* dv = reinterpret_cast<VT*>(dst);
   for (size_t i = tid; i < nv; i += stride) {
     alignas(16) uint8_t q[V];
-    *reinterpret_cast<uint4*>(q) =
-        *reinterpret_cast<const uint4*>(payload + i * V);
+    if (V == 4)
+      *reinterpret_cast<uint32_t*>(q) =
+          *reinterpret_cast<const uint32_t*>(payload + i * 4);
+    else
+      *reinterpret_cast<uint4*>(q) =
+          *reinterpret_cast<const uint4*>(payload + i * 16);
 #pragma unroll
     for (int o = 0; o < NOUT; ++o) {
       VT out;
@@ -439,11 +443,9 @@ __global__ void dequant_reduce_kernel(const uint8_t* __restrict__ in,
   }
   __syncthreads();
 
-  // 16 elements per lane: ONE 16-byte u8 payload load per chunk per
-  // iteration (the CDNA4 coalescing sweet spot — the per-T Vec16 width
-  // would shrink payload loads to 4/8 bytes), then 16*sizeof(T)/16
-  // 16-byte stores out.
-  constexpr int V = 16;
+  // per-dtype width, same tradeoff as dequantize_kernel (measured
+  // r2c9): f32 V=4, f16/bf16 V=16
+  constexpr int V = sizeof(T) == 4 ? 4 : 16;
   constexpr int OV = Vec16<T>::N;       // elems per 16B output vector
   constexpr int NOUT = V / OV;          // output vectors per iteration
   using VT = Vec16<T>;
@@ -458,8 +460,12 @@ __global__ void dequant_reduce_kernel(const uint8_t* __restrict__ in,
     for (int c = 0; c < num_chunks; ++c) {
       const uint8_t* payload = in + (size_t)c * chunk_stride + 32;
       alignas(16) uint8_t q[V];
-      *reinterpret_cast<uint4*>(q) =
-          *reinterpret_cast<const uint4*>(payload + i * V);
+      if (V == 4)
+        *reinterpret_cast<uint32_t*>(q) =
+            *reinterpret_cast<const uint32_t*>(payload + i * 4);
+      else
+        *reinterpret_cast<uint4*>(q) =
+            *reinterpret_cast<const uint4*>(payload + i * 16);
       const float lower = s_lower[c], inv_scale = s_inv_scale[c];
 #pragma unroll
       for (int k = 0; k < V; ++k)
