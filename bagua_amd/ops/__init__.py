@@ -110,9 +110,13 @@ def dequant_reduce(buf: torch.Tensor, flat: torch.Tensor, num_chunks: int,
     reduction inputs on the ByteGrad path). Bitwise-identical to
     decompress_chunked_into + reduce_chunk_inplace: values round through
     flat.dtype between dequantize and f32 accumulation."""
-    if _use_native(buf, flat):
+    if _use_native(buf, flat) and num_chunks <= 64:
         native.lib().dequant_reduce(buf, flat, num_chunks, target_chunk,
                                     average)
+        return
+    if _use_native(buf, flat):  # beyond the kernel's LDS param table
+        decompress_chunked_into(buf, flat, num_chunks)
+        reduce_chunk_inplace(flat, num_chunks, target_chunk, average)
         return
     chunk = flat.numel() // num_chunks
     dec = quant.decompress_chunked(buf, num_chunks, chunk, flat.dtype)

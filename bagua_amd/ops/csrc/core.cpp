@@ -576,9 +576,16 @@ class BucketExecutor {
     // old decompress-then-reduce pair paid ~2 extra bucket passes of
     // HBM traffic for nothing (bitwise-identical result — the kernel
     // rounds through T between dequantize and accumulate)
-    bagua_dequant_reduce_launch(dt, wire, b.flat.data_ptr(), chunk,
-                                stride, n, rank, b.average ? 1 : 0,
-                                stream_);
+    if (n <= 64) {
+      bagua_dequant_reduce_launch(dt, wire, b.flat.data_ptr(), chunk,
+                                  stride, n, rank, b.average ? 1 : 0,
+                                  stream_);
+    } else {  // beyond the kernel's LDS param table: unfused fallback
+      bagua_decompress_launch(dt, wire, b.flat.data_ptr(), chunk, stride,
+                              0, n, stream_);
+      bagua_reduce_chunk_launch(dt, b.flat.data_ptr(), n, rank,
+                                b.average ? 1 : 0, chunk, stream_);
+    }
     bagua_compress_launch(dt, b.flat.data_ptr(), wire, scratch, partials,
                           chunk, stride, n, rank, 1, stream_);
     // in-place allgather of the rank's wire chunk (identity at n==1)
