@@ -400,6 +400,20 @@ class BaguaCommunicator:
         self._run(native, fallback, [tensor])
 
     def alltoall(self, send_tensor, recv_tensor):
+        from .ops import p2p
+
+        if (p2p.enabled() and self.nranks() > 1 and _is_cuda_job()
+                and send_tensor.is_cuda):
+            impl = p2p.get_for_communicator(
+                self, send_tensor.numel() * send_tensor.element_size())
+
+            def native_p2p():
+                impl.alltoall(send_tensor, recv_tensor)
+
+            self._run(native_p2p, lambda: None,
+                      [send_tensor, recv_tensor])
+            return
+
         def native():
             self._native.alltoall(send_tensor, recv_tensor)
 

@@ -56,18 +56,48 @@ def gumbel_rsample(shape, device: torch.device) -> Tensor:
     return gumbel(shape)
 
 
+def _bagua_comm_for_torch_group(group):
+    """BaguaCommunicator for a torch EP group (cached on the group).
+
+    Lazy construction is collective-safe: every EP-group member reaches
+    the MoE layer's first alltoall at the same schedule point, and
+    subset communicators use member-only (local-synchronization) group
+    creation."""
+    from ... import communication
+
+    if not communication.is_initialized():
+        return None
+    if group is None:
+        return communication._get_default_group().get_global_communicator()
+    comm = getattr(group, "_bagua_moe_comm", None)
+    if comm is None:
+        pg = communication.from_torch_group(group)
+        comm = pg.get_global_communicator()
+        group._bagua_moe_comm = comm
+    return comm
+
+
 class _AllToAll(torch.autograd.Function):
     """Autograd alltoall over the EP group
-    (reference: sharded_moe.py:77-90)."""
+    (reference: sharded_moe.py:77-90).
+
+    On GPU with bagua initialized, the exchange runs through the bagua
+    communicator: native RCCL on the group's dedicated comm stream
+    (event-fenced with the compute stream), and the direct one-hop xGMI
+    path when BAGUA_P2P_ALLTOALL=1. Falls back to
+    torch.distributed.all_to_all_single otherwise."""
 
     @staticmethod
     def forward(ctx, group, input: Tensor) -> Tensor:
         ctx.group = group
         input = input.contiguous()
         output = torch.empty_like(input)
-        if group is None or not dist.is_initialized() \
-                or dist.get_world_size(group) == 1:
+        if not dist.is_initialized() or dist.get_world_size(group) == 1:
             output.copy_(input)
+            return output
+        comm = _bagua_comm_for_torch_group(group) if input.is_cuda else None
+        if comm is not None and comm.nranks() > 1:
+            comm.alltoall(input.view(-1), output.view(-1))
         else:
             dist.all_to_all_single(output, input, group=group)
         return output
