@@ -727,9 +727,15 @@ def new_group(ranks: Optional[List[int]] = None, stream=None,
 
 def from_torch_group(group, stream=None) -> BaguaProcessGroup:
     """Convert a torch.distributed group to a BaguaProcessGroup
-    (reference: communication.py:279-310)."""
+    (reference: communication.py:279-310).
+
+    The name must be derived from the RANK SET, not id(group): the name
+    keys the RCCL unique-id exchange in the store, and python object ids
+    differ across ranks (id-based names deadlock every rank on its own
+    key)."""
     ranks = sorted(dist.get_process_group_ranks(group))
-    return new_group(ranks, stream, "from_torch_" + str(id(group)))
+    name = "from_torch_" + "_".join(str(r) for r in ranks)
+    return new_group(ranks, stream, name)
 
 
 def _new_comm_stream():
