@@ -732,10 +732,21 @@ def from_torch_group(group, stream=None) -> BaguaProcessGroup:
     The name must be derived from the RANK SET, not id(group): the name
     keys the RCCL unique-id exchange in the store, and python object ids
     differ across ranks (id-based names deadlock every rank on its own
-    key)."""
+    key). Conversions are cached per rank set so two torch groups over
+    the same ranks share one communicator instead of re-consuming the
+    same unique id."""
     ranks = sorted(dist.get_process_group_ranks(group))
+    key = tuple(ranks)
+    cached = _from_torch_cache.get(key)
+    if cached is not None:
+        return cached
     name = "from_torch_" + "_".join(str(r) for r in ranks)
-    return new_group(ranks, stream, name)
+    pg = new_group(ranks, stream, name)
+    _from_torch_cache[key] = pg
+    return pg
+
+
+_from_torch_cache = {}
 
 
 def _new_comm_stream():
@@ -810,6 +821,7 @@ def deinit_process_group():
         _autotune_server.shutdown()
         _autotune_server = None
     _backends.clear()
+    _from_torch_cache.clear()
     _cached_torch_group.cache_clear()
     _get_rank_mappings.cache_clear()
 
