@@ -46,3 +46,5 @@ from . import contrib  # noqa: F401
 from . import checkpoint  # noqa: F401
 from .parallel import algorithms  # noqa: F401
 from .parallel import moe as model_parallel_moe  # noqa: F401
+from .parallel import moe  # noqa: F401  (reference exports `moe` directly)
+from . import data_parallel  # noqa: F401
