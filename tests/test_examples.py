@@ -38,3 +38,22 @@ def test_moe_example_two_ranks():
                 "examples/moe/main.py", "--steps", "5",
                 "--batch-size", "16"])
     assert "checkpoint saved+restored" in out
+
+
+def test_imagenet_example_two_ranks():
+    out = _run([sys.executable, "-m", "bagua_amd.distributed.launch",
+                "--nproc_per_node", "2", "--master_port",
+                str(find_free_port()),
+                "examples/imagenet/main.py", "--arch", "resnet50",
+                "--epochs", "1", "--batches-per-epoch", "3",
+                "--batch-size", "4"])
+    assert "img/s (whole job)" in out
+
+
+def test_squad_example_two_ranks():
+    out = _run([sys.executable, "-m", "bagua_amd.distributed.launch",
+                "--nproc_per_node", "2", "--master_port",
+                str(find_free_port()),
+                "examples/squad/main.py", "--steps", "3",
+                "--batch-size", "2", "--seq-len", "64"])
+    assert "tokens/s (whole job)" in out
