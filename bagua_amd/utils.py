@@ -79,3 +79,57 @@ class StatisticalAverage:
         if len(self.records) < 2:
             return 0.0
         return self.records[-1][0] - self.records[0][0]
+
+
+def check_contiguous(tensors) -> bool:
+    """True iff the tensors lie back-to-back in one storage
+    (reference: utils.py:51-58)."""
+    data_ptr = None
+    for t in tensors:
+        if data_ptr is not None and t.data_ptr() != data_ptr:
+            return False
+        data_ptr = t.data_ptr() + t.numel() * t.element_size()
+    return True
+
+
+def apply_flattened_call(tensors, call):
+    """Flatten ``tensors`` (same dtype) into one buffer, run ``call`` on
+    it, copy the result back (reference: utils.py:16-29)."""
+    flat = flatten(tensors)
+    call(flat)
+    offset = 0
+    for t in tensors:
+        t.copy_(flat.narrow(0, offset, t.numel()).view_as(t))
+        offset += t.numel()
+
+
+def apply_flattened_call_all(tensors, call):
+    """Group by dtype, then apply_flattened_call per group
+    (reference: utils.py:31-49)."""
+    groups = {}
+    for t in tensors:
+        groups.setdefault(t.dtype, []).append(t)
+    for group in groups.values():
+        apply_flattened_call(group, call)
+
+
+def average_by_removing_extreme_values(raw_scores):
+    """Robust mean of speed samples: drop the warmup third, then
+    iteratively remove >1-sigma outliers (reference: utils.py:94-126).
+    Returns (mean, std, kept_samples)."""
+    import numpy as np
+
+    scores = np.asarray(raw_scores, dtype=float)
+    scores = scores[len(scores) // 3:]
+
+    def weed(x):
+        mean, std = np.mean(x), np.std(x)
+        kept = x[np.abs(x - mean) < std]
+        return kept if len(kept) else x
+
+    scores = weed(scores)
+    for _ in range(10):
+        if np.std(scores) < np.mean(scores):
+            break
+        scores = weed(scores)
+    return float(np.mean(scores)), float(np.std(scores)), scores.tolist()

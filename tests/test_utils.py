@@ -85,3 +85,33 @@ def _worker_trace(rank, nprocs, tmpdir):
         os.path.join(tmpdir, "trace_%d.json" % rank))
     bagua_amd.deinit_process_group()
     return True
+
+
+def test_flattened_call_and_contiguous():
+    import torch
+
+    from bagua_amd.utils import (
+        apply_flattened_call_all,
+        check_contiguous,
+        flatten,
+    )
+
+    ts = [torch.randn(3), torch.randn(5), torch.ones(2, dtype=torch.int64)]
+    ref = [t * 2 if t.is_floating_point() else t * 2 for t in ts]
+    apply_flattened_call_all(ts, lambda flat: flat.mul_(2))
+    for a, b in zip(ts, ref):
+        assert torch.equal(a, b)
+
+    flat = flatten([torch.randn(3), torch.randn(4)])
+    views = [flat[:3], flat[3:]]
+    assert check_contiguous(views)
+    assert not check_contiguous([torch.randn(3), torch.randn(3)])
+
+
+def test_average_by_removing_extreme_values():
+    from bagua_amd.utils import average_by_removing_extreme_values
+
+    samples = [0.0, 0.0, 100.0, 101.0, 99.0, 100.5, 1000.0, 100.2]
+    mean, std, kept = average_by_removing_extreme_values(samples)
+    assert 95.0 < mean < 105.0, mean
+    assert 1000.0 not in kept
