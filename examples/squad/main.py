@@ -58,7 +58,7 @@ def build_model(args, device):
     return create_model("bert-large").to(device), False
 
 
-def batches(args, device, hf):
+def batches(args, device):
     torch.manual_seed(4321 + env.get_rank())
     if args.squad_json and args.model_dir:
         # real data: tokenize locally (no downloads)
@@ -124,7 +124,7 @@ def main():
 
     t0 = time.time()
     tokens = 0
-    for i, batch in enumerate(batches(args, device, hf)):
+    for i, batch in enumerate(batches(args, device)):
         optimizer.zero_grad()
         with amp:
             if hf:
