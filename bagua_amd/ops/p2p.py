@@ -40,14 +40,20 @@ class P2pAllToAll:
         self.rank = rank
         self.nranks = nranks
         self.capacity = capacity
+        from .. import communication
+
         self.impl = native.lib().P2PAlltoAll(
             rank=rank, nranks=nranks, stream=stream_ptr, capacity=capacity)
         store = c10d._get_default_store()
-        key = "bagua_p2p_{}_{}_{}".format(name, epoch, rank)
+        # keyed by BOTH the grow-epoch and the process-group deinit epoch
+        # so stale handle blobs from a previous init can never be read
+        key = "bagua_p2p_{}_{}_{}_{}".format(
+            name, communication._uid_epoch, epoch, rank)
         store.set(key, base64.b64encode(bytes(self.impl.handles())).decode())
         handles = []
         for p in range(nranks):
-            pkey = "bagua_p2p_{}_{}_{}".format(name, epoch, p)
+            pkey = "bagua_p2p_{}_{}_{}_{}".format(
+                name, communication._uid_epoch, epoch, p)
             handles.append(base64.b64decode(store.get(pkey)))
         self.impl.connect(handles)
         logger.info("p2p alltoall connected: %d ranks, %d MiB capacity",
