@@ -174,3 +174,57 @@ def test_native_executor_matches_python_executor(algo):
     assert torch.allclose(a, b, atol=1e-6), (
         "native executor deviates from python executor (max diff %g)"
         % (a - b).abs().max().item())
+
+
+@requires_gpu
+def test_hip_graph_step_capture():
+    """The whole training step (fwd+bwd+bucket schedule+optimizer)
+    must capture into a hipGraph and replay with a sane trajectory
+    (bench.py --hip-graph feature)."""
+    _setup_env()
+    import bagua_amd
+    from bagua_amd.models import MnistNet
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+
+    torch.cuda.set_device(0)
+    bagua_amd.init_process_group()
+    torch.manual_seed(5)
+    model = MnistNet().cuda()
+    optimizer = torch.optim.SGD(model.parameters(), lr=0.01)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm())
+    data = torch.randn(16, 1, 28, 28, device="cuda")
+    target = torch.randint(0, 10, (16,), device="cuda")
+
+    def step():
+        optimizer.zero_grad(set_to_none=False)
+        loss = F.nll_loss(ddp(data), target)
+        loss.backward()
+        optimizer.step()
+        return loss
+
+    for _ in range(3):
+        step()
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for _ in range(3):
+            step()
+    torch.cuda.current_stream().wait_stream(side)
+    torch.cuda.synchronize()
+
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        step()
+    before = torch.cat([p.detach().reshape(-1).clone()
+                        for p in model.parameters()])
+    for _ in range(5):
+        g.replay()
+    torch.cuda.synchronize()
+    after = torch.cat([p.detach().reshape(-1)
+                       for p in model.parameters()])
+    assert torch.isfinite(after).all()
+    assert not torch.equal(before, after), "replay did not train"
