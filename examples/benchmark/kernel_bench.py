@@ -92,6 +92,16 @@ def main():
         p, g, m, v, 10, 1e-3, 0.9, 0.999, 1e-8, 0.0, False), 7 * n * 4)
     results.append(("fused_adam_step", "f32", gbs, us))
 
+    # pure-bf16 flagship path: bf16 params/grads + fp32 master & momentum
+    pb = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    gb = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    master = pb.float()
+    mm = torch.zeros(n, device="cuda")
+    # bytes: read g (2B) + master (4) + m (4); write p (2) + master (4) + m (4)
+    gbs, us = bench(lambda: lib.fused_sgd_mixed_step(
+        pb, gb, master, mm, 0.01, 0.9, 0.0, 1e-4, False, True), 20 * n)
+    results.append(("fused_sgd_mixed", "bf16", gbs, us))
+
     print("%-20s %-5s %10s %10s" % ("kernel", "dtype", "GB/s", "us"))
     for name, dt, gbs, us in results:
         print("%-20s %-5s %10.0f %10.1f" % (name, dt, gbs, us))
