@@ -885,7 +885,8 @@ def _coalesce(tensors, max_bytes):
     group, size = [], 0
     for t in tensors:
         nb = t.numel() * t.element_size()
-        if group and (size + nb > max_bytes or group[0].dtype != t.dtype):
+        if group and (size + nb > max_bytes or group[0].dtype != t.dtype
+                      or group[0].device != t.device):
             yield group
             group, size = [], 0
         group.append(t)

@@ -221,6 +221,7 @@ class BaguaDistributedDataParallel:
                        for p in group["params"]]
         state = optimizer.state_dict()["state"]
         tensors = []
+        cpu_tensors = []
         scalars = {}
         for pid, pstate in sorted(state.items()):
             if (isinstance(pid, int) and pid < len(flat_params)
@@ -228,11 +229,24 @@ class BaguaDistributedDataParallel:
                 continue
             for key, value in sorted(pstate.items()):
                 if isinstance(value, torch.Tensor):
-                    tensors.append(value.data)
+                    # torch Adam keeps its `step` counter as a CPU scalar
+                    # tensor even when the params live on GPU; cat-ing it
+                    # with CUDA state crashes, and RCCL cannot broadcast
+                    # host memory — stage those through the device
+                    if torch.cuda.is_available() and not value.is_cuda:
+                        cpu_tensors.append(value.data)
+                    else:
+                        tensors.append(value.data)
                 else:
                     scalars["{}_{}".format(pid, key)] = value
         if tensors:
             broadcast_coalesced(tensors, src=0, comm=comm)
+        if cpu_tensors:
+            staged = [t.cuda() for t in cpu_tensors]
+            broadcast_coalesced(staged, src=0, comm=comm)
+            torch.cuda.synchronize()
+            for t, s in zip(cpu_tensors, staged):
+                t.copy_(s.cpu())
         if scalars:
             synced = broadcast_object(scalars, src=0, comm=comm)
             for pid, pstate in state.items():

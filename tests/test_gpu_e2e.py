@@ -230,3 +230,40 @@ def test_hip_graph_step_capture():
                        for p in model.parameters()])
     assert torch.isfinite(after).all()
     assert not torch.equal(before, after), "replay did not train"
+
+
+@requires_gpu
+def test_adamw_state_broadcast_on_gpu():
+    """Regression (found by the SQuAD example on hardware): torch Adam
+    keeps its `step` counter as a CPU scalar tensor even for CUDA
+    params; the init optimizer-state broadcast must stage it through the
+    device instead of cat-ing mixed-device tensors."""
+    _setup_env()
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+    from tests.test_algorithms import Net
+
+    torch.cuda.set_device(0)
+    bagua_amd.init_process_group()
+    torch.manual_seed(6)
+    model = Net().cuda()
+    optimizer = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    # materialize state (cpu `step` tensor + cuda moments) BEFORE wrap
+    out = model(torch.randn(8, 8, device="cuda"))
+    out.sum().backward()
+    optimizer.step()
+    optimizer.zero_grad()
+
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer],
+        algorithm=GradientAllReduceAlgorithm())
+    for _ in range(2):
+        loss = F.mse_loss(ddp(torch.randn(8, 8, device="cuda")),
+                          torch.randn(8, 4, device="cuda"))
+        optimizer.zero_grad()
+        loss.backward()
+        optimizer.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
