@@ -64,10 +64,15 @@ def _worker_strategy(rank, nprocs, algo, kwargs):
     ("gradient_allreduce", {}),
     ("bytegrad", {}),
     ("qadam", {}),
+    ("decentralized", {"peer_selection_mode": "all"}),
+    ("low_precision_decentralized", {}),
 ])
 def test_strategy_consensus(algo, kwargs):
     out = run_multi_process(2, _worker_strategy, args=(algo, dict(kwargs)))
-    assert torch.equal(out[0], out[1]), "strategy-wrapped ranks diverged"
+    if algo in ("gradient_allreduce", "bytegrad", "qadam"):
+        assert torch.equal(out[0], out[1]), "strategy ranks diverged"
+    else:  # decentralized family: one local step past the last sync
+        assert torch.allclose(out[0], out[1], atol=0.5)
 
 
 def test_strategy_rejects_missing_qadam_optimizer():

@@ -257,3 +257,42 @@ def test_world8_moe_expert_parallel():
     for r in range(1, WORLD):
         assert not torch.equal(out[0][1], out[r][1]), (
             "experts identical across EP ranks")
+
+
+def _worker_qadam_2x4(rank, nprocs):
+    """QAdam at 2 pseudo-nodes x 4: warmup -> compressed-momentum
+    transition (algorithm re-init) under the hierarchical shape."""
+    os.environ["NODE_RANK"] = str(rank // 4)
+    os.environ["LOCAL_RANK"] = str(rank % 4)
+    os.environ["LOCAL_WORLD_SIZE"] = "4"
+
+    import bagua_amd
+    from bagua_amd.parallel.algorithms import GlobalAlgorithmRegistry
+    from bagua_amd.parallel.algorithms.q_adam import QAdamOptimizer
+    from tests.test_algorithms import Net, _make_data
+
+    bagua_amd.init_process_group()
+    torch.manual_seed(13 + rank)
+    model = Net()
+    optimizer = QAdamOptimizer(model.parameters(), lr=1e-3,
+                               warmup_steps=4)
+    algorithm = GlobalAlgorithmRegistry.get("qadam")(optimizer,
+                                                     hierarchical=True)
+    ddp = bagua_amd.DistributedDataParallel(
+        model, optimizers=[optimizer], algorithm=algorithm)
+    for step in range(9):
+        data, target = _make_data(rank, step)
+        optimizer.zero_grad()
+        loss = F.mse_loss(ddp(data), target)
+        loss.backward()
+        optimizer.step()
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    bagua_amd.deinit_process_group()
+    return flat
+
+
+def test_world8_qadam_hierarchical_2x4():
+    out = run_multi_process(WORLD, _worker_qadam_2x4, timeout=420)
+    for r in range(1, WORLD):
+        assert torch.equal(out[0], out[r]), (
+            "2x4 hierarchical qadam diverged at rank %d" % r)
