@@ -161,3 +161,44 @@ def test_alltoall_v():
         expect = torch.cat([torch.ones(rank + 1) * p
                             for p in range(nprocs)])
         assert torch.allclose(recv, expect)
+
+
+def _worker_reinit_cycle(rank, nprocs):
+    """deinit -> re-init -> train again in ONE process: epoch guards must
+    give the re-created communicators fresh uid/handle store keys and
+    the caches must not leak stale state (multi-model long-lived
+    processes)."""
+    import torch
+    import torch.nn.functional as F
+
+    import bagua_amd
+    from bagua_amd.parallel.algorithms.gradient_allreduce import (
+        GradientAllReduceAlgorithm,
+    )
+    from tests.test_algorithms import Net, _make_data
+
+    flats = []
+    for cycle in range(2):
+        bagua_amd.init_process_group()
+        torch.manual_seed(13 + rank + 100 * cycle)
+        model = Net()
+        optimizer = torch.optim.SGD(model.parameters(), lr=0.05)
+        ddp = bagua_amd.DistributedDataParallel(
+            model, optimizers=[optimizer],
+            algorithm=GradientAllReduceAlgorithm())
+        for step in range(3):
+            data, target = _make_data(rank, step)
+            optimizer.zero_grad()
+            F.mse_loss(ddp(data), target).backward()
+            optimizer.step()
+        flats.append(torch.cat([p.detach().reshape(-1)
+                                for p in model.parameters()]))
+        bagua_amd.deinit_process_group()
+    return flats
+
+
+def test_deinit_reinit_cycle():
+    out = run_multi_process(2, _worker_reinit_cycle)
+    for cycle in range(2):
+        assert torch.equal(out[0][cycle], out[1][cycle]), (
+            "cycle %d diverged after re-init" % cycle)
